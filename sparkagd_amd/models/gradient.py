@@ -1,0 +1,89 @@
+"""Gradient (loss) plug-ins.
+
+Batched equivalents of MLlib 1.3's ``LogisticGradient`` /
+``LeastSquaresGradient`` / ``HingeGradient`` (invoked per example by the
+reference at ``AcceleratedGradientDescent.scala:198``; constructed in
+``AcceleratedGradientDescentSuite.scala:39,251``). Where MLlib folds one
+example at a time into an accumulator, these evaluate a whole GPU shard with
+one fused HIP kernel sequence (margins = A·w, elementwise multiplier,
+grad = A^T·m, loss reduction).
+
+A per-example ``compute`` with MLlib's exact signature is kept for API
+parity and for documentation of the single-example math.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from .. import ops
+
+
+class Gradient:
+    """Computes summed loss and gradient of a loss function over a shard."""
+
+    LOSS_TYPE: int = -1
+
+    def eval(
+        self,
+        shard,
+        w: torch.Tensor,
+        mask: Optional[torch.Tensor] = None,
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Return (grad_sum [d], loss_count float64 [2]) over the local shard.
+
+        ``loss_count[0]`` is the sum of per-example losses; ``loss_count[1]``
+        is the number of (unmasked) examples. Dividing by the globally
+        all-reduced count happens in the optimizer, matching
+        ``AGD.scala:206-207``.
+        """
+        return shard.eval(w, self.LOSS_TYPE, mask)
+
+    # --- MLlib per-example API parity (reference Gradient.compute) ---
+    def compute(
+        self,
+        features: torch.Tensor,
+        label: float,
+        weights: torch.Tensor,
+        cum_gradient: torch.Tensor,
+    ) -> float:
+        """Add one example's gradient into ``cum_gradient`` in place; return its loss.
+
+        Same contract as MLlib's ``Gradient.compute(data, label, weights,
+        cumGradient): Double`` (reference usage ``AGD.scala:198``).
+        """
+        grad_sum, loss_count = ops.reference.dense_eval(
+            features.reshape(1, -1),
+            torch.tensor([label], dtype=torch.float64, device=features.device),
+            weights,
+            self.LOSS_TYPE,
+        )
+        cum_gradient += grad_sum.to(cum_gradient.dtype)
+        return float(loss_count[0])
+
+
+class LogisticGradient(Gradient):
+    """Binary logistic loss; labels in {0, 1}.
+
+    loss_i = y>0 ? log1p(exp(-z)) : log1p(exp(z)),  mult_i = sigmoid(z) - y,
+    with z = <w, x_i> (MLlib writes margin = -z; the algebra is identical).
+    """
+
+    LOSS_TYPE = ops.LOSS_LOGISTIC
+
+
+class LeastSquaresGradient(Gradient):
+    """Squared loss: loss_i = (z - y)^2, mult_i = 2 (z - y)."""
+
+    LOSS_TYPE = ops.LOSS_LEAST_SQUARES
+
+
+class HingeGradient(Gradient):
+    """Hinge loss (linear SVM); labels in {0, 1}, scaled to s = 2y-1 in {-1, 1}.
+
+    loss_i = max(0, 1 - s z), mult_i = -s when s z < 1 else 0.
+    """
+
+    LOSS_TYPE = ops.LOSS_HINGE

@@ -1,0 +1,92 @@
+"""User-layer model trainers (the reference's L6).
+
+In the reference this layer is MLlib's trainer classes (e.g.
+``LogisticRegressionWithSGD``-style wrappers) that construct an optimizer and
+call ``optimize()`` (in-repo example
+``AcceleratedGradientDescentSuite.scala:217-222``). These are the equivalent
+conveniences: a GLM trainer per loss family, returning a fitted linear model
+with ``predict``.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .. import ops
+from ..config import AGDConfig
+from ..optimizer import AcceleratedGradientDescent
+from ..parallel.comm import Communicator
+from .gradient import HingeGradient, LeastSquaresGradient, LogisticGradient
+from .updater import SimpleUpdater, SquaredL2Updater, Updater
+
+
+class LinearModel:
+    """A fitted generalized linear model: weights [d] (+ loss history)."""
+
+    def __init__(self, weights: torch.Tensor, loss_history, link: str):
+        self.weights = weights
+        self.loss_history = list(loss_history)
+        self.link = link
+
+    def margins(self, features: torch.Tensor) -> torch.Tensor:
+        acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+        return features.to(acc) @ self.weights.to(acc)
+
+    def predict(self, features: torch.Tensor) -> torch.Tensor:
+        z = self.margins(features)
+        if self.link == "identity":
+            return z
+        # logistic / hinge: class in {0, 1}
+        return (z > 0).to(torch.float32)
+
+    def predict_proba(self, features: torch.Tensor) -> torch.Tensor:
+        if self.link != "logistic":
+            raise ValueError("predict_proba only for logistic models")
+        return torch.sigmoid(self.margins(features))
+
+
+class _GLMTrainer:
+    GRADIENT_CLS = LogisticGradient
+    LINK = "logistic"
+
+    @classmethod
+    def train(
+        cls,
+        data,
+        num_iterations: int = 100,
+        reg_param: float = 0.0,
+        convergence_tol: float = 1e-4,
+        updater: Optional[Updater] = None,
+        initial_weights: Optional[torch.Tensor] = None,
+        comm: Optional[Communicator] = None,
+        config: Optional[AGDConfig] = None,
+    ) -> LinearModel:
+        cfg = config or AGDConfig()
+        cfg.num_iterations = num_iterations
+        cfg.reg_param = reg_param
+        cfg.convergence_tol = convergence_tol
+        if updater is None:
+            updater = SquaredL2Updater() if reg_param > 0 else SimpleUpdater()
+        opt = AcceleratedGradientDescent(cls.GRADIENT_CLS(), updater, cfg, comm)
+        if initial_weights is None:
+            wdtype = torch.float64 if data.device.type == "cpu" else torch.float32
+            initial_weights = torch.zeros(data.d, device=data.device, dtype=wdtype)
+        w = opt.optimize(data, initial_weights)
+        return LinearModel(w, opt.loss_history, cls.LINK)
+
+
+class LogisticRegressionWithAGD(_GLMTrainer):
+    GRADIENT_CLS = LogisticGradient
+    LINK = "logistic"
+
+
+class LinearRegressionWithAGD(_GLMTrainer):
+    GRADIENT_CLS = LeastSquaresGradient
+    LINK = "identity"
+
+
+class SVMWithAGD(_GLMTrainer):
+    GRADIENT_CLS = HingeGradient
+    LINK = "hinge"

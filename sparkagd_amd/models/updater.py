@@ -1,0 +1,71 @@
+"""Updater (proximal-step) plug-ins.
+
+Semantics of MLlib 1.3's ``SimpleUpdater`` / ``L1Updater`` /
+``SquaredL2Updater`` (invoked by the reference at
+``AcceleratedGradientDescent.scala:215-220``): one proximal gradient step plus
+the regularization value at the new point. MLlib internally rescales the step
+as stepSize/sqrt(iter); the AGD driver defeats that by passing iter=1
+(``AGD.scala:218-219``) — we reproduce the rescaling here (host-side scalar)
+so ``runMiniBatch`` gets the decaying schedule and AGD gets the raw step.
+
+On GPU the step runs as one fused HIP prox kernel per updater type that also
+emits the regularization scalar via a block reduction (no weight broadcast is
+ever needed: every rank executes the identical update deterministically).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Tuple
+
+import torch
+
+from .. import ops
+
+
+class Updater:
+    """Updates weights given a gradient step; returns the new weights and the
+    value of the regularization term at the new weights."""
+
+    PROX_KIND: int = -1
+
+    def compute(
+        self,
+        weights_old: torch.Tensor,
+        gradient: torch.Tensor,
+        step_size: float,
+        iter: int,  # noqa: A002 - MLlib parameter name
+        reg_param: float,
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """MLlib contract: step = step_size / sqrt(iter); returns (w_new, reg_value).
+
+        ``reg_value`` is a float64 scalar tensor (device-resident on GPU).
+        """
+        this_step = step_size / math.sqrt(iter)
+        return ops.prox(self.PROX_KIND, weights_old, gradient, this_step, reg_param)
+
+    def reg_value(self, weights: torch.Tensor, reg_param: float) -> torch.Tensor:
+        """Regularization value at ``weights`` (the reference's step=0 trick,
+        ``AGD.scala:305``: applyProjector(x, g, 0) returns c_x)."""
+        zero_grad = torch.zeros_like(weights)
+        _, reg = ops.prox(self.PROX_KIND, weights, zero_grad, 0.0, reg_param)
+        return reg
+
+
+class SimpleUpdater(Updater):
+    """w' = w - step*g; no regularization."""
+
+    PROX_KIND = ops.PROX_SIMPLE
+
+
+class L1Updater(Updater):
+    """Gradient step then soft-threshold prox: w'_i = sign(w1_i)*max(0, |w1_i| - step*lambda);
+    reg = lambda * ||w'||_1."""
+
+    PROX_KIND = ops.PROX_L1
+
+
+class SquaredL2Updater(Updater):
+    """w' = w*(1 - step*lambda) - step*g; reg = (lambda/2) ||w'||^2."""
+
+    PROX_KIND = ops.PROX_SQUARED_L2

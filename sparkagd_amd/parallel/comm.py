@@ -1,0 +1,113 @@
+"""Distributed communication: RCCL all-reduce over xGMI.
+
+The MI355X-native replacement for the reference's Spark data plane
+(``AcceleratedGradientDescent.scala:192-208``): where the reference does a
+torrent broadcast of the weight vector (C2) plus a depth-2 ``treeAggregate``
+of (loss, grad, count) over Akka/Netty TCP (C1), here:
+
+* the **broadcast is eliminated** — every rank holds (x, z, theta, L) and runs
+  the prox/Nesterov update kernels on identical all-reduced inputs, so the
+  weights at y are already resident and bit-identical on every rank (RCCL
+  all-reduce delivers the same bytes to all ranks). A broadcast remains only
+  for initial weights and for debug divergence checks.
+* the tree-reduce becomes ONE ``all_reduce(sum)`` of the grad_sum buffer plus
+  one 2-element float64 all-reduce of (loss_sum, count) per evaluation,
+  via ``torch.distributed`` whose "nccl" backend IS RCCL on ROCm. On a single
+  8-GPU node RCCL's in-node algorithms use all 7 point-to-point xGMI links
+  (≈153 GB/s each); at d=1e6 fp32 the 4 MB message is latency-dominated and
+  far below the cost of one data pass.
+
+On CPU (tests/CI) the identical code runs over the gloo backend; with no
+process group initialized every call is a no-op (world size 1).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+
+class Communicator:
+    """Thin wrapper over a torch.distributed process group (or none)."""
+
+    def __init__(self, group: Optional[object] = None):
+        self.group = group
+        self._active = dist.is_available() and dist.is_initialized()
+
+    @property
+    def world_size(self) -> int:
+        if not self._active:
+            return 1
+        return dist.get_world_size(self.group)
+
+    @property
+    def rank(self) -> int:
+        if not self._active:
+            return 0
+        return dist.get_rank(self.group)
+
+    def allreduce_(self, t: torch.Tensor) -> torch.Tensor:
+        """In-place sum all-reduce (no-op at world size 1)."""
+        if self._active and self.world_size > 1:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.group)
+        return t
+
+    def allreduce_eval_(self, grad_sum: torch.Tensor, loss_count: torch.Tensor):
+        """All-reduce one evaluation's (grad_sum, (loss_sum, count)).
+
+        Two collectives because grad is f32 (or the weight dtype) while the
+        loss/count pair stays f64 for the backtracking state machine's
+        cancellation-sensitive f_y≈f_x test (``AGD.scala:272-278``).
+        """
+        self.allreduce_(grad_sum)
+        self.allreduce_(loss_count)
+        return grad_sum, loss_count
+
+    def broadcast_(self, t: torch.Tensor, src: int = 0) -> torch.Tensor:
+        """Used only for initial weights and debug divergence checks (the
+        per-iteration weight broadcast of the reference, ``AGD.scala:193``,
+        is eliminated by deterministic replicated updates)."""
+        if self._active and self.world_size > 1:
+            dist.broadcast(t, src=src, group=self.group)
+        return t
+
+    def barrier(self) -> None:
+        if self._active and self.world_size > 1:
+            if dist.get_backend(self.group) == "nccl" and torch.cuda.is_available():
+                dist.barrier(group=self.group, device_ids=[torch.cuda.current_device()])
+            else:
+                dist.barrier(group=self.group)
+
+    def check_replicated(self, t: torch.Tensor, rtol: float = 0.0) -> bool:
+        """Debug divergence check: is ``t`` identical (or within rtol) across
+        ranks? Implemented as broadcast-from-0 + local compare."""
+        if not self._active or self.world_size == 1:
+            return True
+        ref = t.clone()
+        self.broadcast_(ref, src=0)
+        if rtol == 0.0:
+            ok = bool(torch.equal(ref, t))
+        else:
+            ok = bool(torch.allclose(ref, t, rtol=rtol, atol=0.0))
+        flag = torch.tensor([0.0 if ok else 1.0], dtype=torch.float64, device=t.device)
+        self.allreduce_(flag)
+        return bool(flag.item() == 0.0)
+
+
+def init_from_env(backend: Optional[str] = None) -> Communicator:
+    """Initialize torch.distributed from torchrun env vars (RANK/WORLD_SIZE/
+    LOCAL_RANK/MASTER_*) and bind this process to its GPU. Safe to call when
+    not launched under torchrun (returns a world-size-1 Communicator)."""
+    if "RANK" not in os.environ or "WORLD_SIZE" not in os.environ:
+        return Communicator()
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if backend == "nccl" and torch.cuda.is_available():
+            local_rank = int(os.environ.get("LOCAL_RANK", os.environ["RANK"]))
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
+        dist.init_process_group(backend=backend)
+    return Communicator()

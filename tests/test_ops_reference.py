@@ -1,0 +1,157 @@
+"""Unit tests of the reference (oracle) ops against hand-written NumPy math.
+
+These pin the MLlib-1.3 loss/updater semantics so the
+HIP kernels can later be validated against `ops.reference` transitively.
+"""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from sparkagd_amd import ops
+from sparkagd_amd.models.gradient import (
+    HingeGradient,
+    LeastSquaresGradient,
+    LogisticGradient,
+)
+from sparkagd_amd.models.updater import L1Updater, SimpleUpdater, SquaredL2Updater
+from sparkagd_amd.data import DenseShard, CSRShard, generate_csr_problem
+
+
+def _np_eval(A, y, w, loss_type):
+    z = A @ w
+    if loss_type == ops.LOSS_LOGISTIC:
+        mult = 1.0 / (1.0 + np.exp(-z)) - y
+        loss = np.where(y > 0, np.log1p(np.exp(-np.abs(z))) + np.maximum(-z, 0),
+                        np.log1p(np.exp(-np.abs(z))) + np.maximum(z, 0))
+    elif loss_type == ops.LOSS_LEAST_SQUARES:
+        mult = 2.0 * (z - y)
+        loss = (z - y) ** 2
+    else:
+        s = 2.0 * y - 1.0
+        viol = s * z < 1.0
+        mult = np.where(viol, -s, 0.0)
+        loss = np.maximum(0.0, 1.0 - s * z)
+    return A.T @ mult, loss.sum()
+
+
+@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES, ops.LOSS_HINGE])
+def test_dense_eval_matches_numpy(loss_type):
+    rng = np.random.default_rng(0)
+    n, d = 257, 13
+    A = rng.normal(size=(n, d))
+    y = (rng.normal(size=n) > 0).astype(np.float64)
+    if loss_type == ops.LOSS_LEAST_SQUARES:
+        y = rng.normal(size=n)
+    w = rng.normal(size=d)
+    grad_np, loss_np = _np_eval(A, y, w, loss_type)
+
+    grad, lc = ops.reference.dense_eval(
+        torch.from_numpy(A), torch.from_numpy(y), torch.from_numpy(w), loss_type
+    )
+    np.testing.assert_allclose(grad.numpy(), grad_np, rtol=1e-10)
+    assert abs(float(lc[0]) - loss_np) < 1e-8 * max(1.0, abs(loss_np))
+    assert float(lc[1]) == n
+
+
+@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES, ops.LOSS_HINGE])
+def test_dense_eval_mask(loss_type):
+    rng = np.random.default_rng(1)
+    n, d = 100, 7
+    A = rng.normal(size=(n, d))
+    y = (rng.normal(size=n) > 0).astype(np.float64)
+    w = rng.normal(size=d)
+    mask = (rng.random(n) < 0.5)
+    grad_np, loss_np = _np_eval(A[mask], y[mask], w, loss_type)
+    grad, lc = ops.reference.dense_eval(
+        torch.from_numpy(A), torch.from_numpy(y), torch.from_numpy(w), loss_type,
+        mask=torch.from_numpy(mask.astype(np.uint8)),
+    )
+    np.testing.assert_allclose(grad.numpy(), grad_np, rtol=1e-10)
+    assert abs(float(lc[0]) - loss_np) < 1e-8 * max(1.0, abs(loss_np))
+    assert float(lc[1]) == mask.sum()
+
+
+def test_csr_eval_matches_dense():
+    shard, _ = generate_csr_problem(n=300, d=50, nnz_per_row=5, seed=3)
+    w = torch.randn(50, dtype=torch.float32)
+    grad_c, lc_c = shard.eval(w, ops.LOSS_LOGISTIC)
+    # densify
+    A = torch.zeros(300, 50)
+    for i in range(300):
+        for k in range(int(shard.rowptr[i]), int(shard.rowptr[i + 1])):
+            A[i, int(shard.col[k])] += float(shard.val[k])
+    grad_d, lc_d = ops.reference.dense_eval(A, shard.labels, w, ops.LOSS_LOGISTIC)
+    torch.testing.assert_close(grad_c, grad_d, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(lc_c, lc_d, rtol=1e-8, atol=1e-8)
+
+
+def test_gradient_compute_per_example_parity():
+    """The MLlib per-example Gradient.compute contract (AGD.scala:198)."""
+    rng = np.random.default_rng(2)
+    d = 5
+    for g_cls, label in [(LogisticGradient, 1.0), (LeastSquaresGradient, 0.37), (HingeGradient, 0.0)]:
+        g = g_cls()
+        x = torch.from_numpy(rng.normal(size=d))
+        w = torch.from_numpy(rng.normal(size=d))
+        cum = torch.zeros(d, dtype=torch.float64)
+        loss = g.compute(x, label, w, cum)
+        grad_np, loss_np = _np_eval(x.numpy().reshape(1, -1), np.array([label]), w.numpy(), g.LOSS_TYPE)
+        np.testing.assert_allclose(cum.numpy(), grad_np, rtol=1e-10)
+        assert abs(loss - loss_np) < 1e-10 * max(1.0, abs(loss_np))
+
+
+def test_updaters_mllib_semantics():
+    rng = np.random.default_rng(4)
+    d = 11
+    w = rng.normal(size=d)
+    g = rng.normal(size=d)
+    wt, gt = torch.from_numpy(w), torch.from_numpy(g)
+
+    # SimpleUpdater: w - (s/sqrt(iter)) g, reg 0
+    w2, reg = SimpleUpdater().compute(wt, gt, 0.7, 4, 0.3)
+    np.testing.assert_allclose(w2.numpy(), w - (0.7 / 2.0) * g, rtol=1e-12)
+    assert float(reg) == 0.0
+
+    # L1Updater: soft threshold, reg = lam*||w'||_1
+    lam, s0, it = 0.3, 0.7, 9
+    s = s0 / math.sqrt(it)
+    w1 = w - s * g
+    expected = np.sign(w1) * np.maximum(np.abs(w1) - lam * s, 0.0)
+    w2, reg = L1Updater().compute(wt, gt, s0, it, lam)
+    np.testing.assert_allclose(w2.numpy(), expected, rtol=1e-12)
+    assert abs(float(reg) - lam * np.abs(expected).sum()) < 1e-10
+
+    # SquaredL2Updater: w(1-s*lam) - s*g, reg = lam/2 ||w'||^2
+    w2, reg = SquaredL2Updater().compute(wt, gt, s0, it, lam)
+    expected = w * (1 - s * lam) - s * g
+    np.testing.assert_allclose(w2.numpy(), expected, rtol=1e-12)
+    assert abs(float(reg) - 0.5 * lam * (expected**2).sum()) < 1e-10
+
+
+def test_fused_scalars():
+    rng = np.random.default_rng(5)
+    d = 33
+    x, y, gy, xo = (rng.normal(size=d) for _ in range(4))
+    out = ops.reference.fused_scalars(*(torch.from_numpy(v) for v in (x, y, gy, xo))).numpy()
+    np.testing.assert_allclose(out[0], ((x - y) ** 2).sum(), rtol=1e-12)
+    np.testing.assert_allclose(out[1], ((x - y) * gy).sum(), rtol=1e-12)
+    np.testing.assert_allclose(out[2], (x * x).sum(), rtol=1e-12)
+    np.testing.assert_allclose(out[3], ((x - xo) ** 2).sum(), rtol=1e-12)
+    np.testing.assert_allclose(out[4], (gy * (x - xo)).sum(), rtol=1e-12)
+
+
+def test_axpby():
+    x = torch.randn(17, dtype=torch.float64)
+    y = torch.randn(17, dtype=torch.float64)
+    out = ops.reference.axpby(0.3, x, -1.7, y)
+    torch.testing.assert_close(out, 0.3 * x - 1.7 * y)
+
+
+def test_shard_properties():
+    sh = DenseShard(torch.randn(10, 4), torch.zeros(10))
+    assert sh.n == 10 and sh.d == 4 and sh.nbytes > 0
+    c, _ = generate_csr_problem(20, 8, 3, seed=0)
+    assert c.n == 20 and c.d == 8 and c.nnz == 60
