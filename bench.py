@@ -1,0 +1,210 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: distributed accelerated gradient descent on dense
+logistic regression, d = 10^6, bf16 shards, one process per MI355X over RCCL.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+(for N > 1 the driver launches this under torch.distributed.run, one rank per
+GPU; RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* come from the env). W untimed warmup
+AGD iterations, then EXACTLY K timed iterations bracketed by a barrier +
+torch.cuda.synchronize() on both sides; elapsed is the MAX over ranks; rank 0
+prints ONE JSON line.
+
+Metric: examples/sec — examples processed through the gradient pipeline
+(evaluations x rows) per second, aggregated over all ranks. Each AGD iteration
+performs 2 full-data evaluations in the default backtracking configuration
+(eval at y + the accepted backtracking eval at x; the reference's third
+TFOCS-validation pass, AGD.scala:302-307, is off — loss history reuses the
+accepted f_x). Weak scaling: rows-per-GPU fixed as N grows.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import math
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from sparkagd_amd import (  # noqa: E402
+    AGDConfig,
+    LogisticGradient,
+    LeastSquaresGradient,
+    HingeGradient,
+    SimpleUpdater,
+    SquaredL2Updater,
+    generate_dense_problem,
+    run,
+)
+from sparkagd_amd import ops  # noqa: E402
+from sparkagd_amd.data import generate_csr_problem  # noqa: E402
+from sparkagd_amd.parallel.comm import init_from_env  # noqa: E402
+
+BASELINE_METRIC = "examples/sec + iters-to-ε, logistic regression d=10^6 at 1/2/4/8 MI355X"
+
+LOSSES = {
+    "logistic": (ops.LOSS_LOGISTIC, LogisticGradient),
+    "lsq": (ops.LOSS_LEAST_SQUARES, LeastSquaresGradient),
+    "hinge": (ops.LOSS_HINGE, HingeGradient),
+}
+
+
+class CountingGradient:
+    def __init__(self, inner):
+        self.inner = inner
+        self.n_evals = 0
+
+    def eval(self, shard, w, mask=None):
+        self.n_evals += 1
+        return self.inner.eval(shard, w, mask)
+
+
+def sync(device: torch.device) -> None:
+    if device.type == "cuda":
+        torch.cuda.synchronize(device)
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--rows", type=int, default=16384, help="rows per GPU (weak scaling)")
+    p.add_argument("--d", type=int, default=1_000_000)
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "f32"])
+    p.add_argument("--loss", type=str, default="logistic", choices=list(LOSSES))
+    p.add_argument("--reg", type=float, default=0.0)
+    p.add_argument("--csr", action="store_true", help="CSR-sparse shard instead of dense")
+    p.add_argument("--nnz-per-row", type=int, default=64)
+    p.add_argument("--eps", type=float, default=1e-3, help="relative loss-improvement epsilon for iters-to-eps")
+    args = p.parse_args()
+
+    comm = init_from_env()
+    rank, world = comm.rank, comm.world_size
+    if torch.cuda.is_available():
+        device = torch.device("cuda", torch.cuda.current_device())
+        dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+        wdtype = torch.float32
+    else:  # CPU smoke fallback (the real bench runs on MI355X)
+        device = torch.device("cpu")
+        dtype = torch.float32
+        wdtype = torch.float64
+        args.rows = min(args.rows, 2048)
+        args.d = min(args.d, 512)
+
+    loss_type, grad_cls = LOSSES[args.loss]
+
+    t_gen0 = time.perf_counter()
+    if args.csr:
+        shard, _w_true = generate_csr_problem(
+            args.rows, args.d, args.nnz_per_row, seed=1234 + rank * 7,
+            loss_type=loss_type, device=device,
+        )
+    else:
+        shard, _w_true = generate_dense_problem(
+            args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
+            device=device, dtype=dtype,
+        )
+    sync(device)
+    t_gen = time.perf_counter() - t_gen0
+
+    gradient = CountingGradient(grad_cls())
+    updater = SquaredL2Updater() if args.reg > 0 else SimpleUpdater()
+    w0 = torch.zeros(args.d, device=device, dtype=wdtype)
+
+    state = {"t0": 0.0, "t1": 0.0, "e0": 0, "e1": 0, "timed_iters": 0}
+    total_iters = args.warmup + args.steps
+
+    def hook(n_iter: int):
+        if n_iter == args.warmup:
+            comm.barrier()
+            sync(device)
+            state["t0"] = time.perf_counter()
+            state["e0"] = gradient.n_evals
+        if n_iter == total_iters:
+            comm.barrier()
+            sync(device)
+            state["t1"] = time.perf_counter()
+            state["e1"] = gradient.n_evals
+            state["timed_iters"] = n_iter - args.warmup
+            return "stop"
+        return None
+
+    if args.warmup == 0:
+        comm.barrier()
+        sync(device)
+        state["t0"] = time.perf_counter()
+
+    weights, hist = run(
+        shard, gradient, updater,
+        0.0,                      # convergence_tol: never stop early in a bench
+        total_iters, args.reg, w0,
+        1.0, math.inf, 0.5, 0.9, True,
+        loss_history_mode="backtrack",
+        comm=comm,
+        iteration_hook=hook,
+    )
+
+    elapsed_local = state["t1"] - state["t0"]
+    evals = state["e1"] - state["e0"]
+    timed_iters = state["timed_iters"] or args.steps
+
+    # elapsed = MAX over ranks
+    el = torch.tensor([elapsed_local], dtype=torch.float64, device=device)
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.all_reduce(el, op=dist.ReduceOp.MAX)
+    elapsed = float(el[0])
+
+    global_rows = args.rows * world
+    examples = global_rows * evals  # evals identical on all ranks (replicated control flow)
+    value = examples / elapsed if elapsed > 0 else float("nan")
+
+    # iters-to-eps on the recorded loss history
+    iters_to_eps = None
+    for i in range(1, len(hist)):
+        if abs(hist[i] - hist[i - 1]) < args.eps * max(abs(hist[i]), 1e-30):
+            iters_to_eps = i + 1
+            break
+
+    if rank == 0:
+        out = {
+            "metric": BASELINE_METRIC,
+            "value": value,
+            "unit": "examples/s",
+            "n_gpus": world,
+            "steps": timed_iters,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed * 1000.0 / timed_iters,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if device.type == "cuda" else "f32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{'csr' if args.csr else 'dense'}_{args.loss}_regression",
+                "d": args.d,
+                "rows_per_gpu": args.rows,
+                "global_rows": global_rows,
+                "parallelism": f"dp{world}",
+                "evals_per_step": evals / max(timed_iters, 1),
+                "weights_dtype": str(wdtype).replace("torch.", ""),
+                "loss_final": hist[-1] if hist else None,
+                "iters_to_eps": iters_to_eps,
+                "eps": args.eps,
+                "gen_seconds": round(t_gen, 3),
+                "shard_gb": round(shard.nbytes / 2**30, 3),
+            },
+        }
+        print(json.dumps(out), flush=True)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

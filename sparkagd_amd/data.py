@@ -159,7 +159,7 @@ def generate_dense_problem(
     loss_type: int = ops.LOSS_LOGISTIC,
     device: str | torch.device = "cpu",
     dtype: torch.dtype = torch.float32,
-    chunk_rows: int = 65536,
+    chunk_rows: Optional[int] = None,
     label_noise: float = 0.1,
 ) -> Tuple[DenseShard, torch.Tensor]:
     """Random-init dense problem with planted weights, generated in row chunks
@@ -170,6 +170,9 @@ def generate_dense_problem(
     with z = X @ w_true and w_true ~ N(0, 1/sqrt(d)).
     """
     dev = torch.device(device)
+    if chunk_rows is None:
+        # bound the fp32 staging buffer to ~1 GB regardless of d
+        chunk_rows = max(64, min(65536, (1 << 28) // max(d, 1)))
     gen = torch.Generator(device=dev).manual_seed(seed)
     w_true = torch.randn(d, generator=gen, device=dev, dtype=torch.float32) / math.sqrt(d)
     feats = torch.empty((n, d), device=dev, dtype=dtype)

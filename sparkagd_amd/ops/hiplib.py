@@ -1,0 +1,218 @@
+"""ctypes binding to the CDNA4 kernel library (csrc/libagd_hip.so).
+
+The .so is a plain C-ABI HIP library built directly by hipcc for gfx950 —
+no hipify, no torch C++ headers, no pybind — loaded with ctypes and called
+with raw device pointers plus torch's current HIP stream. The build lives
+in-tree (``sparkagd_amd/csrc/libagd_hip.so``) so it travels with the repo
+snapshot; ``__graft_entry__.build()`` produces it.
+
+Determinism notes:
+* the dense A^T·m path writes private partial slabs per (row-block,
+  column-slab) workgroup and reduces them in a fixed order => bitwise
+  reproducible gradients across runs (and therefore across ranks given
+  identical inputs).
+* the CSR A^T·m path uses fp32 atomics (scatter); its summation order is
+  non-deterministic, which perturbs the gradient at the fp32-rounding level.
+  Tests compare it against the fp64 torch oracle with tolerance; a
+  deterministic CSC-transpose variant is the planned alternative
+  (SURVEY.md §5 'Race detection').
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Optional, Tuple
+
+import torch
+
+_DTYPE_CODE = {torch.bfloat16: 0, torch.float32: 1, torch.float64: 2}
+_ACC_DTYPE = {torch.bfloat16: torch.float32, torch.float32: torch.float32, torch.float64: torch.float64}
+
+_lib: Optional[ctypes.CDLL] = None
+
+SO_PATH = os.path.join(os.path.dirname(os.path.abspath(__file__)), "..", "csrc", "libagd_hip.so")
+
+
+def so_path() -> str:
+    return os.path.abspath(SO_PATH)
+
+
+def load() -> ctypes.CDLL:
+    global _lib
+    if _lib is not None:
+        return _lib
+    path = so_path()
+    if not os.path.exists(path):
+        raise FileNotFoundError(f"HIP kernel library not found at {path}")
+    lib = ctypes.CDLL(path)
+
+    lib.agd_last_error.restype = ctypes.c_char_p
+    lib.agd_version.restype = ctypes.c_int
+    lib.agd_dense_rowblocks.restype = ctypes.c_longlong
+    lib.agd_dense_rowblocks.argtypes = [ctypes.c_longlong, ctypes.c_longlong, ctypes.c_int]
+    lib.agd_margin_slabs.restype = ctypes.c_int
+    lib.agd_margin_slabs.argtypes = [ctypes.c_longlong, ctypes.c_longlong, ctypes.c_int]
+
+    P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
+    lib.agd_dense_eval.restype = I
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, P]
+    lib.agd_csr_eval.restype = I
+    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P]
+    lib.agd_axpby.restype = I
+    lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
+    lib.agd_prox.restype = I
+    lib.agd_prox.argtypes = [I, P, P, D, D, P, P, LL, I, P]
+    lib.agd_fused_scalars.restype = I
+    lib.agd_fused_scalars.argtypes = [P, P, P, P, P, LL, I, P]
+    lib.agd_dot_diff.restype = I
+    lib.agd_dot_diff.argtypes = [P, P, P, P, P, LL, I, P]
+
+    _lib = lib
+    return lib
+
+
+def _check(rc: int) -> None:
+    if rc != 0:
+        raise RuntimeError(f"agd HIP kernel error rc={rc}: {_lib.agd_last_error().decode()}")
+
+
+def _stream(t: torch.Tensor) -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream(t.device).cuda_stream)
+
+
+def _ptr(t: Optional[torch.Tensor]) -> Optional[ctypes.c_void_p]:
+    if t is None:
+        return None
+    return ctypes.c_void_p(t.data_ptr())
+
+
+def _prep_mask(mask: Optional[torch.Tensor], device) -> Optional[torch.Tensor]:
+    if mask is None:
+        return None
+    if mask.dtype != torch.uint8:
+        mask = mask.to(torch.uint8)
+    return mask.contiguous().to(device)
+
+
+def dense_eval(
+    features: torch.Tensor,
+    labels: torch.Tensor,
+    w: torch.Tensor,
+    loss_type: int,
+    mask: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    lib = load()
+    assert features.is_cuda and features.is_contiguous() and features.ndim == 2
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    acc = _ACC_DTYPE[features.dtype]
+    if w.dtype != acc:
+        raise TypeError(f"weights dtype {w.dtype} must be {acc} for {features.dtype} features")
+    w = w.contiguous()
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, features.device)
+
+    dev = features.device
+    grad = torch.empty(d, dtype=acc, device=dev)
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    margins = torch.empty(n, dtype=acc, device=dev)
+    mult = torch.empty(n, dtype=acc, device=dev)
+    n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
+    part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
+    n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype))
+
+    rc = lib.agd_dense_eval(
+        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
+        n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
+        _ptr(part), n_rb, loss_type, n_slabs, _stream(features),
+    )
+    _check(rc)
+    return grad, loss_count
+
+
+def csr_eval(
+    rowptr: torch.Tensor,
+    col: torch.Tensor,
+    val: torch.Tensor,
+    labels: torch.Tensor,
+    w: torch.Tensor,
+    loss_type: int,
+    mask: Optional[torch.Tensor] = None,
+    d: Optional[int] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    lib = load()
+    assert val.is_cuda and val.dtype == torch.float32
+    if w.dtype != torch.float32:
+        raise TypeError("CSR path requires float32 weights")
+    n = rowptr.numel() - 1
+    d = d if d is not None else w.numel()
+    rowptr = rowptr.contiguous()
+    col = col.contiguous()
+    if rowptr.dtype != torch.int32:
+        rowptr = rowptr.to(torch.int32)
+    if col.dtype != torch.int32:
+        col = col.to(torch.int32)
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, val.device)
+
+    dev = val.device
+    grad = torch.zeros(d, dtype=torch.float32, device=dev)
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    margins = torch.empty(n, dtype=torch.float32, device=dev)
+    mult = torch.empty(n, dtype=torch.float32, device=dev)
+
+    rc = lib.agd_csr_eval(
+        _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask),
+        _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
+        _ptr(margins), _ptr(mult), loss_type, _stream(val),
+    )
+    _check(rc)
+    return grad, loss_count
+
+
+_VEC_DTYPE = {torch.float32: 1, torch.float64: 2}
+
+
+def axpby(a: float, x: torch.Tensor, b: float, y: torch.Tensor, out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    lib = load()
+    if out is None:
+        out = torch.empty_like(x)
+    rc = lib.agd_axpby(float(a), _ptr(x.contiguous()), float(b), _ptr(y.contiguous()),
+                       _ptr(out), x.numel(), _VEC_DTYPE[x.dtype], _stream(x))
+    _check(rc)
+    return out
+
+
+def prox(kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float) -> Tuple[torch.Tensor, torch.Tensor]:
+    lib = load()
+    out = torch.empty_like(w)
+    reg = torch.zeros((), dtype=torch.float64, device=w.device)
+    rc = lib.agd_prox(kind, _ptr(w.contiguous()), _ptr(g.contiguous()), float(step),
+                      float(lam), _ptr(out), _ptr(reg), w.numel(), _VEC_DTYPE[w.dtype], _stream(w))
+    _check(rc)
+    return out, reg
+
+
+def fused_scalars(x: torch.Tensor, y: torch.Tensor, g_y: torch.Tensor, x_old: torch.Tensor) -> torch.Tensor:
+    lib = load()
+    out = torch.zeros(5, dtype=torch.float64, device=x.device)
+    rc = lib.agd_fused_scalars(_ptr(x.contiguous()), _ptr(y.contiguous()),
+                               _ptr(g_y.contiguous()), _ptr(x_old.contiguous()),
+                               _ptr(out), x.numel(), _VEC_DTYPE[x.dtype], _stream(x))
+    _check(rc)
+    return out
+
+
+def dot_diff(x: torch.Tensor, y: torch.Tensor, g_x: torch.Tensor, g_y: torch.Tensor) -> torch.Tensor:
+    lib = load()
+    out = torch.zeros((), dtype=torch.float64, device=x.device)
+    rc = lib.agd_dot_diff(_ptr(x.contiguous()), _ptr(y.contiguous()),
+                          _ptr(g_x.contiguous()), _ptr(g_y.contiguous()),
+                          _ptr(out), x.numel(), _VEC_DTYPE[x.dtype], _stream(x))
+    _check(rc)
+    return out

@@ -82,6 +82,7 @@ def run(
     checkpoint_path: Optional[str] = None,
     checkpoint_every: int = 0,
     resume_from: Optional[str] = None,
+    iteration_hook=None,
 ) -> Tuple[torch.Tensor, List[float]]:
     """Run accelerated proximal gradient descent.
 
@@ -89,6 +90,9 @@ def run(
     ``AcceleratedGradientDescent.run`` (``AGD.scala:177-189``), returning
     ``(weights, loss_history)``. Keyword-only extras are new capabilities
     (metrics, checkpoint/resume, loss-history mode — SURVEY.md §5).
+    ``iteration_hook(n_iter)`` runs at the end of each completed iteration;
+    returning the string "stop" ends the loop (used by bench.py to bracket
+    exactly K timed steps).
     """
     comm = comm or Communicator()
     backtrack_tol = 1e-10
@@ -232,6 +236,8 @@ def run(
             )
 
         if broke:
+            break
+        if iteration_hook is not None and iteration_hook(n_iter) == "stop":
             break
 
     logger.info(
