@@ -128,50 +128,111 @@ __device__ __forceinline__ void loadW(const TA* __restrict__ p, TACC (&out)[W]) 
   }
 }
 
+// Vector load of W accumulator-typed elements (16-B chunks; the address is
+// W-element aligned by construction in the margins kernel).
+template <typename TACC, int W>
+__device__ __forceinline__ void loadAcc(const TACC* __restrict__ p, TACC (&out)[W]) {
+  constexpr int EPC = 16 / sizeof(TACC);
+  if constexpr (W >= EPC) {
+#pragma unroll
+    for (int ch = 0; ch < W / EPC; ++ch) {
+      if constexpr (sizeof(TACC) == 4) {
+        using f32x4 = __attribute__((ext_vector_type(4))) float;
+        f32x4 v = *(const f32x4*)(p + ch * 4);
+#pragma unroll
+        for (int k = 0; k < 4; ++k) out[ch * 4 + k] = v[k];
+      } else {
+        using f64x2 = __attribute__((ext_vector_type(2))) double;
+        f64x2 v = *(const f64x2*)(p + ch * 2);
+#pragma unroll
+        for (int k = 0; k < 2; ++k) out[ch * 2 + k] = v[k];
+      }
+    }
+  } else {
+#pragma unroll
+    for (int k = 0; k < W; ++k) out[k] = p[k];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // K1a: dense margins  z[r] = <A[r,:], w>
 //
-// One wave per (row, column-slab) task, 16 B per lane per step, wave-shuffle
-// reduction per row. n_slabs > 1 (fat-d / thin-n shards) accumulates with
-// fp32/fp64 atomics into pre-zeroed margins.
+// One wave per (row-group, column-slab) task: R rows share each 16-B-per-lane
+// load of w, so at fat d (where w does not fit L1/L2 and would otherwise be
+// re-streamed per row, 2x the A-traffic at bf16) the w bytes per A byte drop
+// by R. Wave-shuffle reduction per row. n_slabs > 1 (thin-n / fat-d shards)
+// writes per-slab partial dots into part[s*n + r]; the multiplier kernel sums
+// the slab axis => deterministic (no atomics anywhere on the dense path).
 // ---------------------------------------------------------------------------
+
+#define MARGIN_ROWS 4  // R: rows per wave (w-load amortization)
 
 template <typename TA, typename TACC, int W>
 __global__ __launch_bounds__(BLOCK) void k_dense_margins(
     const TA* __restrict__ A, const TACC* __restrict__ w, ll n, ll d,
-    ll slab_w, int n_slabs, TACC* __restrict__ margins) {
+    ll slab_w, int n_slabs, TACC* __restrict__ part) {
+  constexpr int R = MARGIN_ROWS;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
   const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
-  const ll n_tasks = n * n_slabs;
+  const ll n_rg = (n + R - 1) / R;
+  const ll n_tasks = n_rg * n_slabs;
   for (ll t = wave_gid; t < n_tasks; t += n_waves) {
-    const ll r = t / n_slabs;
-    const ll s = t - r * n_slabs;
+    const ll rg = t / n_slabs;
+    const ll s = t - rg * n_slabs;
+    const ll r0 = rg * R;
+    const int nr = (int)((r0 + R <= n) ? R : (n - r0));
     const ll c_lo = s * slab_w;
     const ll c_hi = (c_lo + slab_w < d) ? c_lo + slab_w : d;
-    const TA* __restrict__ row = A + r * d;
-    TACC acc = (TACC)0;
+    const TA* __restrict__ row0 = A + r0 * d;
+    TACC acc[R];
+#pragma unroll
+    for (int j = 0; j < R; ++j) acc[j] = (TACC)0;
     ll c = c_lo + (ll)lane * W;
-    for (; c + W <= c_hi; c += (ll)WAVE * W) {
-      TACC v[W];
-      loadW<TA, TACC, W>(row + c, v);
+    if (nr == R) {  // full row group (hot path)
+      for (; c + W <= c_hi; c += (ll)WAVE * W) {
+        TACC wv[W];
+        loadAcc<TACC, W>(w + c, wv);
 #pragma unroll
-      for (int k = 0; k < W; ++k) acc += v[k] * w[c + k];
-    }
-    if (c < c_hi) {
+        for (int j = 0; j < R; ++j) {
+          TACC v[W];
+          loadW<TA, TACC, W>(row0 + (ll)j * d + c, v);
 #pragma unroll
-      for (int k = 0; k < W; ++k)
-        if (c + k < c_hi) {
-          TACC v[1];
-          loadW<TA, TACC, 1>(row + c + k, v);
-          acc += v[0] * w[c + k];
+          for (int k = 0; k < W; ++k) acc[j] += v[k] * wv[k];
         }
+      }
+      if (c < c_hi) {
+#pragma unroll
+        for (int k = 0; k < W; ++k)
+          if (c + k < c_hi) {
+            const TACC wk = w[c + k];
+#pragma unroll
+            for (int j = 0; j < R; ++j) {
+              TACC v[1];
+              loadW<TA, TACC, 1>(row0 + (ll)j * d + c + k, v);
+              acc[j] += v[0] * wk;
+            }
+          }
+      }
+    } else {  // tail row group
+      for (; c < c_hi; c += (ll)WAVE * W) {
+#pragma unroll
+        for (int k = 0; k < W; ++k)
+          if (c + k < c_hi) {
+            const TACC wk = w[c + k];
+            for (int j = 0; j < nr; ++j) {
+              TACC v[1];
+              loadW<TA, TACC, 1>(row0 + (ll)j * d + c + k, v);
+              acc[j] += v[0] * wk;
+            }
+          }
+      }
     }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) {
-      if (n_slabs == 1) margins[r] = acc;
-      else atomicAdd(&margins[r], acc);
+#pragma unroll
+    for (int j = 0; j < R; ++j) {
+      acc[j] = wave_reduce_sum(acc[j]);
+      if (lane == 0 && j < nr) part[s * n + r0 + j] = acc[j];
     }
   }
 }
@@ -205,7 +266,7 @@ __device__ __forceinline__ float softplus<float>(float t) {
 template <typename TACC>
 __global__ __launch_bounds__(BLOCK) void k_multiplier(
     const TACC* __restrict__ margins, const float* __restrict__ labels,
-    const unsigned char* __restrict__ mask, int loss_type, ll n,
+    const unsigned char* __restrict__ mask, int loss_type, ll n, int n_slabs,
     TACC* __restrict__ mult, double* __restrict__ loss_count) {
   double lsum = 0.0, cnt = 0.0;
   const ll stride = (ll)gridDim.x * BLOCK;
@@ -214,7 +275,8 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
       mult[i] = (TACC)0;
       continue;
     }
-    const TACC z = margins[i];
+    TACC z = margins[i];
+    for (int s = 1; s < n_slabs; ++s) z += margins[(ll)s * n + i];
     const TACC y = (TACC)labels[i];
     TACC m, l;
     if (loss_type == LOSS_LOGISTIC) {
@@ -481,15 +543,17 @@ extern "C" long long agd_dense_rowblocks(long long n, long long d, int a_dtype) 
   return n_rb;
 }
 
-// Margin-pass column slabs (>1 only for thin-n / fat-d shards, where row
-// parallelism alone cannot fill 256 CUs).
+// Margin-pass column slabs (>1 only for thin-n / fat-d shards, where
+// row-group parallelism alone cannot fill 256 CUs). Callers size the margins
+// workspace as n_slabs * n accumulators.
 extern "C" int agd_margin_slabs(long long n, long long d, int a_dtype) {
   const ll target_waves = 16384;
-  if (n >= target_waves) return 1;
+  const ll n_rg = (n + MARGIN_ROWS - 1) / MARGIN_ROWS;
+  if (n_rg >= target_waves) return 1;
   const int w = pick_w(a_dtype, d);
   const ll min_slab = (ll)WAVE * w * 4;  // keep >= 4 vector iterations per slab
   ll max_slabs = (d + min_slab - 1) / min_slab;
-  ll want = (target_waves + n - 1) / n;
+  ll want = (target_waves + n_rg - 1) / n_rg;
   ll s = want < max_slabs ? want : max_slabs;
   if (s < 1) s = 1;
   if (s > 1024) s = 1024;
@@ -516,11 +580,9 @@ static int dense_eval_t(const void* A, const float* labels,
     slab_w = ((slab_w + align - 1) / align) * align;
     n_slabs = (int)((d + slab_w - 1) / slab_w);
   }
-  if (n_slabs > 1)
-    HIP_CHECK(hipMemsetAsync(margins, 0, n * sizeof(TACC), stream));
 
   {
-    const ll tasks = n * n_slabs;
+    const ll tasks = ((n + MARGIN_ROWS - 1) / MARGIN_ROWS) * n_slabs;
     const int grid = grid_for(tasks, WAVES_PER_BLOCK);
     hipLaunchKernelGGL((k_dense_margins<TA, TACC, W>), dim3(grid), dim3(BLOCK),
                        0, stream, a, wp, n, d, slab_w, n_slabs, margins);
@@ -528,7 +590,8 @@ static int dense_eval_t(const void* A, const float* labels,
   {
     const int grid = grid_for(n, BLOCK);
     hipLaunchKernelGGL((k_multiplier<TACC>), dim3(grid), dim3(BLOCK), 0, stream,
-                       margins, labels, mask, loss_type, n, mult, loss_count);
+                       margins, labels, mask, loss_type, n, n_slabs, mult,
+                       loss_count);
   }
   {
     const ll cols_per_block = (ll)BLOCK * W;
@@ -596,7 +659,7 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
     const int grid = grid_for(n, BLOCK);
     hipLaunchKernelGGL((k_multiplier<float>), dim3(grid), dim3(BLOCK), 0, s,
                        margins, (const float*)labels,
-                       (const unsigned char*)mask, loss_type, n, mult,
+                       (const unsigned char*)mask, loss_type, n, 1, mult,
                        (double*)loss_count);
   }
   {

@@ -118,11 +118,11 @@ def dense_eval(
     dev = features.device
     grad = torch.empty(d, dtype=acc, device=dev)
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
-    margins = torch.empty(n, dtype=acc, device=dev)
+    n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype))
+    margins = torch.empty(n_slabs * n, dtype=acc, device=dev)
     mult = torch.empty(n, dtype=acc, device=dev)
     n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
     part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
-    n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype))
 
     rc = lib.agd_dense_eval(
         _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
