@@ -395,6 +395,23 @@ __global__ __launch_bounds__(BLOCK) void k_csr_margins(
   }
 }
 
+// Deterministic CSR transpose gradient via a CSC copy of the shard:
+// grad[c] = sum_{k in column c} val[k] * m[row[k]], one thread per column
+// (mean nnz/column is small at d >= 1e6; m rides L2/LLC). No atomics =>
+// bitwise reproducible; also the faster path when columns are not skewed.
+__global__ __launch_bounds__(BLOCK) void k_csc_grad(
+    const int* __restrict__ colptr, const int* __restrict__ row,
+    const float* __restrict__ val, const float* __restrict__ mult, ll d,
+    float* __restrict__ grad) {
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll c = (ll)blockIdx.x * BLOCK + threadIdx.x; c < d; c += stride) {
+    const int k_lo = colptr[c], k_hi = colptr[c + 1];
+    float acc = 0.f;
+    for (int k = k_lo; k < k_hi; ++k) acc += val[k] * mult[row[k]];
+    grad[c] = acc;
+  }
+}
+
 __global__ __launch_bounds__(BLOCK) void k_csr_grad(
     const int* __restrict__ rowptr, const int* __restrict__ col,
     const float* __restrict__ val, const float* __restrict__ mult, ll n,
@@ -635,15 +652,20 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
   return 2;
 }
 
-// CSR applySmooth: margins -> multiplier/loss -> atomic A^T·m.
-// loss_count AND grad_out must be zeroed by the caller (grad is accumulated).
+// CSR applySmooth: margins -> multiplier/loss -> A^T·m.
+// Two transpose-gradient paths: atomic scatter over the CSR arrays
+// (csc_* == nullptr), or the deterministic CSC gather when the caller
+// provides the column-sorted copy (colptr [d+1], cscrow/cscval [nnz]).
+// loss_count must be zeroed by the caller; grad_out additionally so on the
+// atomic path (it is accumulated there, overwritten on the CSC path).
 extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val,
                             const void* labels, const void* mask, const void* w,
                             long long n, long long nnz, long long d,
                             void* grad_out, void* loss_count, void* margins_ws,
-                            void* mult_ws, int loss_type, void* stream) {
+                            void* mult_ws, int loss_type,
+                            const void* csc_colptr, const void* csc_row,
+                            const void* csc_val, void* stream) {
   (void)nnz;
-  (void)d;
   hipStream_t s = (hipStream_t)stream;
   const int* rp = (const int*)rowptr;
   const int* ci = (const int*)col;
@@ -662,7 +684,12 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
                        (const unsigned char*)mask, loss_type, n, 1, mult,
                        (double*)loss_count);
   }
-  {
+  if (csc_colptr != nullptr) {
+    const int grid = grid_for(d, BLOCK);
+    hipLaunchKernelGGL(k_csc_grad, dim3(grid), dim3(BLOCK), 0, s,
+                       (const int*)csc_colptr, (const int*)csc_row,
+                       (const float*)csc_val, mult, d, (float*)grad_out);
+  } else {
     const int grid = grid_for(n, WAVES_PER_BLOCK);
     hipLaunchKernelGGL(k_csr_grad, dim3(grid), dim3(BLOCK), 0, s, rp, ci, v,
                        mult, n, (float*)grad_out);

@@ -74,7 +74,7 @@ class CSRShard:
     kind = "csr"
 
     def __init__(self, rowptr: torch.Tensor, col: torch.Tensor, val: torch.Tensor,
-                 labels: torch.Tensor, d: int):
+                 labels: torch.Tensor, d: int, deterministic: bool = True):
         self.rowptr = rowptr.contiguous()
         self.col = col.contiguous()
         self.val = val.contiguous()
@@ -82,6 +82,11 @@ class CSRShard:
         if self.labels.dtype not in (torch.float32, torch.float64):
             self.labels = self.labels.to(torch.float32)
         self._d = int(d)
+        # deterministic=True builds a CSC copy of the shard at construction
+        # (2x nnz memory) so the A^T·m pass is a gather instead of an fp32
+        # atomic scatter: bitwise-reproducible gradients (SURVEY.md §5,
+        # 'Race detection'). deterministic=False keeps the atomic path.
+        self.csc = self._build_csc() if deterministic else None
 
     @property
     def n(self) -> int:
@@ -106,9 +111,24 @@ class CSRShard:
                 + self.val.numel() * self.val.element_size()
                 + self.labels.numel() * self.labels.element_size())
 
+    def _build_csc(self):
+        """Column-sorted (CSC) copy: colptr [d+1] i32, row [nnz] i32, val [nnz] f32."""
+        n = self.rowptr.numel() - 1
+        counts = torch.diff(self.rowptr.to(torch.int64))
+        rows = torch.repeat_interleave(
+            torch.arange(n, device=self.col.device, dtype=torch.int32), counts
+        )
+        order = torch.argsort(self.col.to(torch.int64), stable=True)
+        csc_row = rows[order].contiguous()
+        csc_val = self.val[order].contiguous()
+        colcounts = torch.bincount(self.col.to(torch.int64), minlength=self._d)
+        colptr = torch.zeros(self._d + 1, dtype=torch.int64, device=self.col.device)
+        torch.cumsum(colcounts, dim=0, out=colptr[1:])
+        return colptr.to(torch.int32).contiguous(), csc_row, csc_val
+
     def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None):
         return ops.csr_eval(self.rowptr, self.col, self.val, self.labels, w,
-                            loss_type, mask, self._d)
+                            loss_type, mask, self._d, csc=self.csc)
 
 
 # ---------------------------------------------------------------------------

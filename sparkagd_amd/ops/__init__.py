@@ -92,9 +92,11 @@ def csr_eval(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     d: Optional[int] = None,
+    csc=None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(val):
-        return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask, d)
+        return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, csc)
+    # the torch reference (sparse_csr @ / .t() @) is already deterministic
     return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d)
 
 

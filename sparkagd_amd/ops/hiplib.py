@@ -58,7 +58,7 @@ def load() -> ctypes.CDLL:
     lib.agd_dense_eval.restype = I
     lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, P]
     lib.agd_csr_eval.restype = I
-    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P]
+    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, P]
     lib.agd_axpby.restype = I
     lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
     lib.agd_prox.restype = I
@@ -142,6 +142,7 @@ def csr_eval(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     d: Optional[int] = None,
+    csc: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     assert val.is_cuda and val.dtype == torch.float32
@@ -161,7 +162,12 @@ def csr_eval(
     mask = _prep_mask(mask, val.device)
 
     dev = val.device
-    grad = torch.zeros(d, dtype=torch.float32, device=dev)
+    if csc is None:
+        grad = torch.zeros(d, dtype=torch.float32, device=dev)  # atomic path accumulates
+        cp = cr = cv = None
+    else:
+        grad = torch.empty(d, dtype=torch.float32, device=dev)  # CSC path overwrites
+        cp, cr, cv = csc
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
     margins = torch.empty(n, dtype=torch.float32, device=dev)
     mult = torch.empty(n, dtype=torch.float32, device=dev)
@@ -169,7 +175,8 @@ def csr_eval(
     rc = lib.agd_csr_eval(
         _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask),
         _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
-        _ptr(margins), _ptr(mult), loss_type, _stream(val),
+        _ptr(margins), _ptr(mult), loss_type,
+        _ptr(cp), _ptr(cr), _ptr(cv), _stream(val),
     )
     _check(rc)
     return grad, loss_count

@@ -83,6 +83,7 @@ def run(
     checkpoint_every: int = 0,
     resume_from: Optional[str] = None,
     iteration_hook=None,
+    check_replication_every: int = 0,
 ) -> Tuple[torch.Tensor, List[float]]:
     """Run accelerated proximal gradient descent.
 
@@ -115,9 +116,21 @@ def run(
         loss_history = list(state["loss_history"])
         start_iter = state["iter"] + 1
 
+    eval_state = {"n": 0, "seconds": 0.0}
+
+    def apply_smooth(v, mask=None):
+        t0 = time.perf_counter()
+        out = _apply_smooth(data, gradient, comm, v, mask)
+        # _apply_smooth ends with the host fetch of (loss, count), so this
+        # wall segment covers the kernels + all-reduce for the evaluation.
+        eval_state["n"] += 1
+        eval_state["seconds"] += time.perf_counter() - t0
+        return out
+
     broke = False
     for n_iter in range(start_iter, num_iterations + 1):
         t_iter0 = time.perf_counter()
+        eval_n0, eval_s0 = eval_state["n"], eval_state["seconds"]
         # Auslender and Teboulle's accelerated method (AGD.scala:237-255).
         x_old, z_old = x, z
         L_old = L
@@ -135,7 +148,7 @@ def run(
             # after a restart gives theta = 1 (AGD.scala:248).
             theta = 2.0 / (1.0 + math.sqrt(1.0 + 4.0 * (L / L_old) / (theta_old * theta_old)))
             y = ops.axpby(1.0 - theta, x_old, theta, z_old)
-            f_y, g_y, _count = _apply_smooth(data, gradient, comm, y)
+            f_y, g_y, _count = apply_smooth(y)
             step = 1.0 / (theta * L)
             z, _ = updater.compute(z_old, g_y, step, 1, reg_param)
             x = ops.axpby(1.0 - theta, x_old, theta, z)
@@ -151,7 +164,7 @@ def run(
             if xy_sq == 0.0:
                 break
 
-            f_x, g_x, _ = _apply_smooth(data, gradient, comm, x)
+            f_x, g_x, _ = apply_smooth(x)
             f_x_bt = f_x
             if backtrack_simple:
                 q_x = f_y + float(scal[1]) + 0.5 * L * xy_sq
@@ -177,7 +190,7 @@ def run(
         # extra full-data pass at x (TFOCS validation); 'backtrack' reuses the
         # accepted backtracking evaluation; 'none' records f_y + c_y.
         if loss_history_mode == "exact":
-            f_x2, _g_x2, _ = _apply_smooth(data, gradient, comm, x)
+            f_x2, _g_x2, _ = apply_smooth(x)
             c_x = float(updater.reg_value(x, reg_param))
             loss_history.append(f_x2 + c_x)
         elif loss_history_mode == "backtrack" and f_x_bt is not None:
@@ -208,6 +221,16 @@ def run(
             backtrack_simple = True
             restarted = True
 
+        if (
+            check_replication_every > 0
+            and n_iter % check_replication_every == 0
+            and not comm.check_replicated(x)
+        ):
+            raise RuntimeError(
+                f"rank divergence detected at iteration {n_iter}: the "
+                "replicated weight state is no longer identical across ranks"
+            )
+
         if metrics is not None:
             metrics.log(
                 iter=n_iter,
@@ -219,6 +242,8 @@ def run(
                 restarted=restarted,
                 norm_dx=norm_dx,
                 iter_seconds=time.perf_counter() - t_iter0,
+                n_evals=eval_state["n"] - eval_n0,
+                eval_seconds=eval_state["seconds"] - eval_s0,
             )
 
         if (
@@ -260,6 +285,7 @@ def run_mini_batch(
     comm: Optional[Communicator] = None,
     seed: int = 42,
     metrics=None,
+    step_schedule: str = "sqrt",
 ) -> Tuple[torch.Tensor, List[float]]:
     """Mini-batch (S)GD — the golden-baseline optimizer.
 
@@ -272,7 +298,16 @@ def run_mini_batch(
 
     On GPU the sampling is a seeded per-shard mask pushed into the gradient
     kernels (masked rows contribute neither loss nor gradient).
+
+    ``step_schedule``: 'sqrt' is MLlib's step_size/sqrt(i) (the default and
+    the parity mode); 'constant' uses step_size; 'linear' is the
+    strong-convexity (Pegasos-style) schedule 1/(reg_param * i) for
+    L2-regularized objectives (BASELINE.md hinge/SVM config).
     """
+    if step_schedule not in ("sqrt", "constant", "linear"):
+        raise ValueError("step_schedule must be sqrt|constant|linear")
+    if step_schedule == "linear" and reg_param <= 0:
+        raise ValueError("linear (strong-convexity) schedule needs reg_param > 0")
     comm = comm or Communicator()
     w = initial_weights.clone()
     history: List[float] = []
@@ -297,7 +332,11 @@ def run_mini_batch(
         if batch_size > 0:
             history.append(loss_sum / batch_size + reg_val)
             grad_sum.div_(batch_size)
-            w, reg_t = updater.compute(w, grad_sum, step_size, i, reg_param)
+            if step_schedule == "sqrt":
+                w, reg_t = updater.compute(w, grad_sum, step_size, i, reg_param)
+            else:
+                s = step_size if step_schedule == "constant" else 1.0 / (reg_param * i)
+                w, reg_t = updater.compute(w, grad_sum, s, 1, reg_param)
             reg_val = float(reg_t)
             if metrics is not None:
                 metrics.log(iter=i, loss=history[-1], batch_size=batch_size)

@@ -110,6 +110,23 @@ def test_csr_eval_matches_reference(loss_type):
     torch.testing.assert_close(lc_h, lc_r, rtol=1e-5, atol=1e-6)
 
 
+def test_csr_csc_deterministic_vs_atomic():
+    """The CSC-gather A^T·m equals the atomic-scatter path (tolerance) and is
+    bitwise reproducible run-to-run (SURVEY.md §5 race-detection cross-check)."""
+    from sparkagd_amd.ops import hiplib
+
+    shard, _ = generate_csr_problem(n=30000, d=40000, nnz_per_row=24, seed=17, device=DEV)
+    g = torch.Generator(device=DEV).manual_seed(5)
+    w = torch.randn(40000, generator=g, device=DEV, dtype=torch.float32) * 0.1
+    args = (shard.rowptr, shard.col, shard.val, shard.labels, w, ops.LOSS_LOGISTIC, None, 40000)
+    g_atomic, lc_a = hiplib.csr_eval(*args, csc=None)
+    g_csc1, lc_c = hiplib.csr_eval(*args, csc=shard.csc)
+    g_csc2, _ = hiplib.csr_eval(*args, csc=shard.csc)
+    assert torch.equal(g_csc1, g_csc2)  # deterministic
+    torch.testing.assert_close(g_csc1, g_atomic, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(lc_a, lc_c, rtol=0, atol=0)
+
+
 @pytest.mark.parametrize("dtype", [torch.float32, torch.float64])
 def test_axpby(dtype):
     from sparkagd_amd.ops import hiplib

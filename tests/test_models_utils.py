@@ -1,0 +1,105 @@
+"""Tests for the user-layer trainers (L6), metrics, config, and the
+mini-batch step schedules."""
+
+import json
+import math
+
+import pytest
+import torch
+
+from sparkagd_amd import AGDConfig, HingeGradient, SquaredL2Updater, run_mini_batch
+from sparkagd_amd.data import generate_dense_problem, generate_logistic_data
+from sparkagd_amd.models.trainers import (
+    LinearRegressionWithAGD,
+    LogisticRegressionWithAGD,
+    SVMWithAGD,
+)
+from sparkagd_amd.utils.metrics import JsonlMetrics, NullMetrics
+from sparkagd_amd import ops
+
+
+def test_logistic_trainer_accuracy():
+    shard, w_true = generate_dense_problem(2000, 16, seed=0, dtype=torch.float64)
+    model = LogisticRegressionWithAGD.train(shard, num_iterations=30)
+    pred = model.predict(shard.features)
+    acc = float((pred == shard.labels.to(pred.dtype)).double().mean())
+    assert acc > 0.9
+    proba = model.predict_proba(shard.features)
+    assert float(proba.min()) >= 0 and float(proba.max()) <= 1
+
+
+def test_linear_trainer_recovers_weights():
+    shard, w_true = generate_dense_problem(
+        4000, 8, seed=1, loss_type=ops.LOSS_LEAST_SQUARES, dtype=torch.float64
+    )
+    model = LinearRegressionWithAGD.train(shard, num_iterations=60, convergence_tol=1e-10)
+    err = float(torch.norm(model.weights - w_true.to(model.weights.dtype)) / torch.norm(w_true))
+    assert err < 0.15
+
+
+def test_svm_trainer_separates():
+    shard, _ = generate_dense_problem(
+        2000, 12, seed=2, loss_type=ops.LOSS_HINGE, dtype=torch.float64
+    )
+    model = SVMWithAGD.train(shard, num_iterations=40, reg_param=0.01)
+    pred = model.predict(shard.features)
+    acc = float((pred == shard.labels.to(pred.dtype)).double().mean())
+    assert acc > 0.85
+
+
+def test_minibatch_step_schedules():
+    shard, _ = generate_dense_problem(
+        2000, 12, seed=3, loss_type=ops.LOSS_HINGE, dtype=torch.float64
+    )
+    w0 = torch.zeros(12, dtype=torch.float64)
+    for sched in ("sqrt", "constant", "linear"):
+        w, hist = run_mini_batch(
+            shard, HingeGradient(), SquaredL2Updater(), 0.5, 20, 0.05, 1.0, w0,
+            step_schedule=sched,
+        )
+        assert hist[-1] < hist[0], sched
+    with pytest.raises(ValueError):
+        run_mini_batch(shard, HingeGradient(), SquaredL2Updater(), 0.5, 5, 0.0,
+                       1.0, w0, step_schedule="linear")
+
+
+def test_jsonl_metrics(tmp_path):
+    p = str(tmp_path / "m.jsonl")
+    with JsonlMetrics(p, rank=0) as m:
+        m.log(iter=1, loss=0.5)
+        m.log(iter=2, loss=0.25, n_evals=2)
+    rows = [json.loads(line) for line in open(p)]
+    assert rows[0]["iter"] == 1 and rows[1]["n_evals"] == 2
+    # non-zero rank writes nothing
+    p2 = str(tmp_path / "m2.jsonl")
+    with JsonlMetrics(p2, rank=1) as m:
+        m.log(iter=1)
+    import os
+    assert not os.path.exists(p2)
+    NullMetrics().log(x=1)
+
+
+def test_metrics_wired_into_run(tmp_path):
+    import sparkagd_amd
+
+    data = generate_logistic_data(2.0, -1.5, 1000, seed=5)
+    w0 = torch.zeros(2, dtype=torch.float64)
+    p = str(tmp_path / "it.jsonl")
+    with JsonlMetrics(p) as m:
+        sparkagd_amd.run(
+            data, sparkagd_amd.LogisticGradient(), SquaredL2Updater(),
+            1e-12, 5, 0.1, w0, 1.0, math.inf, 0.5, 0.9, True, metrics=m,
+        )
+    rows = [json.loads(line) for line in open(p)]
+    assert len(rows) == 5
+    assert all("loss" in r and "L" in r and "eval_seconds" in r and "n_evals" in r for r in rows)
+    assert rows[0]["n_evals"] >= 2
+
+
+def test_config_json_roundtrip():
+    c = AGDConfig(convergence_tol=1e-6, Lexact=math.inf, beta=0.7)
+    c2 = AGDConfig.from_json(c.to_json())
+    assert c2 == c
+    c.beta = -1
+    with pytest.raises(ValueError):
+        c.validate()

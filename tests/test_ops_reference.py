@@ -88,6 +88,21 @@ def test_csr_eval_matches_dense():
     torch.testing.assert_close(lc_c, lc_d, rtol=1e-8, atol=1e-8)
 
 
+def test_csc_build_matches_csr():
+    """CSRShard's CSC copy re-expresses exactly the same matrix."""
+    shard, _ = generate_csr_problem(n=200, d=60, nnz_per_row=7, seed=8)
+    colptr, row, val = shard.csc
+    A1 = torch.zeros(200, 60, dtype=torch.float64)
+    for i in range(200):
+        for k in range(int(shard.rowptr[i]), int(shard.rowptr[i + 1])):
+            A1[i, int(shard.col[k])] += float(shard.val[k])
+    A2 = torch.zeros(200, 60, dtype=torch.float64)
+    for c in range(60):
+        for k in range(int(colptr[c]), int(colptr[c + 1])):
+            A2[int(row[k]), c] += float(val[k])
+    torch.testing.assert_close(A1, A2)
+
+
 def test_gradient_compute_per_example_parity():
     """The MLlib per-example Gradient.compute contract (AGD.scala:198)."""
     rng = np.random.default_rng(2)
