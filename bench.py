@@ -57,11 +57,13 @@ LOSSES = {
 class CountingGradient:
     def __init__(self, inner):
         self.inner = inner
-        self.n_evals = 0
+        self.n_evals = 0   # loss evaluations (each touches every example)
+        self.n_passes = 0  # full data passes (grad eval = 2: A·w and A^T·m)
 
-    def eval(self, shard, w, mask=None):
+    def eval(self, shard, w, mask=None, need_grad=True):
         self.n_evals += 1
-        return self.inner.eval(shard, w, mask)
+        self.n_passes += 2 if need_grad else 1
+        return self.inner.eval(shard, w, mask, need_grad)
 
 
 def sync(device: torch.device) -> None:
@@ -117,7 +119,7 @@ def main() -> int:
     updater = SquaredL2Updater() if args.reg > 0 else SimpleUpdater()
     w0 = torch.zeros(args.d, device=device, dtype=wdtype)
 
-    state = {"t0": 0.0, "t1": 0.0, "e0": 0, "e1": 0, "timed_iters": 0}
+    state = {"t0": 0.0, "t1": 0.0, "e0": 0, "e1": 0, "p0": 0, "p1": 0, "timed_iters": 0}
     total_iters = args.warmup + args.steps
 
     def hook(n_iter: int):
@@ -126,11 +128,13 @@ def main() -> int:
             sync(device)
             state["t0"] = time.perf_counter()
             state["e0"] = gradient.n_evals
+            state["p0"] = gradient.n_passes
         if n_iter == total_iters:
             comm.barrier()
             sync(device)
             state["t1"] = time.perf_counter()
             state["e1"] = gradient.n_evals
+            state["p1"] = gradient.n_passes
             state["timed_iters"] = n_iter - args.warmup
             return "stop"
         return None
@@ -152,6 +156,7 @@ def main() -> int:
 
     elapsed_local = state["t1"] - state["t0"]
     evals = state["e1"] - state["e0"]
+    passes = state["p1"] - state["p0"]
     timed_iters = state["timed_iters"] or args.steps
 
     # elapsed = MAX over ranks
@@ -194,6 +199,7 @@ def main() -> int:
                 "global_rows": global_rows,
                 "parallelism": f"dp{world}",
                 "evals_per_step": evals / max(timed_iters, 1),
+                "data_passes_per_step": passes / max(timed_iters, 1),
                 "weights_dtype": str(wdtype).replace("torch.", ""),
                 "loss_final": hist[-1] if hist else None,
                 "iters_to_eps": iters_to_eps,

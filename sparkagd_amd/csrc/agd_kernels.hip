@@ -582,7 +582,7 @@ static int dense_eval_t(const void* A, const float* labels,
                         const unsigned char* mask, const void* w, ll n, ll d,
                         void* grad_out, double* loss_count, void* margins_ws,
                         void* mult_ws, void* part_ws, ll n_rb, int loss_type,
-                        int n_slabs, hipStream_t stream) {
+                        int n_slabs, int need_grad, hipStream_t stream) {
   const TA* a = (const TA*)A;
   const TACC* wp = (const TACC*)w;
   TACC* margins = (TACC*)margins_ws;
@@ -610,17 +610,19 @@ static int dense_eval_t(const void* A, const float* labels,
                        margins, labels, mask, loss_type, n, n_slabs, mult,
                        loss_count);
   }
-  {
-    const ll cols_per_block = (ll)BLOCK * W;
-    const ll n_cs = (d + cols_per_block - 1) / cols_per_block;
-    const int grid = grid_for(n_rb * n_cs, 1);
-    hipLaunchKernelGGL((k_dense_grad<TA, TACC, W>), dim3(grid), dim3(BLOCK), 0,
-                       stream, a, mult, n, d, n_rb, part);
-  }
-  if (n_rb > 1) {
-    const int grid = grid_for(d, BLOCK);
-    hipLaunchKernelGGL((k_grad_reduce<TACC>), dim3(grid), dim3(BLOCK), 0,
-                       stream, part, n_rb, d, grad);
+  if (need_grad) {
+    {
+      const ll cols_per_block = (ll)BLOCK * W;
+      const ll n_cs = (d + cols_per_block - 1) / cols_per_block;
+      const int grid = grid_for(n_rb * n_cs, 1);
+      hipLaunchKernelGGL((k_dense_grad<TA, TACC, W>), dim3(grid), dim3(BLOCK),
+                         0, stream, a, mult, n, d, n_rb, part);
+    }
+    if (n_rb > 1) {
+      const int grid = grid_for(d, BLOCK);
+      hipLaunchKernelGGL((k_grad_reduce<TACC>), dim3(grid), dim3(BLOCK), 0,
+                         stream, part, n_rb, d, grad);
+    }
   }
   HIP_CHECK(hipGetLastError());
   return 0;
@@ -634,19 +636,19 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
                               long long d, void* grad_out, void* loss_count,
                               void* margins_ws, void* mult_ws, void* part_ws,
                               long long n_rb, int loss_type, int n_slabs,
-                              void* stream) {
+                              int need_grad, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const float* lab = (const float*)labels;
   const unsigned char* msk = (const unsigned char*)mask;
   double* lc = (double*)loss_count;
   const int W = pick_w(a_dtype, d);
   switch (a_dtype * 10 + (W > 1 ? 1 : 0)) {
-    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, s);
-    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, s);
-    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, s);
-    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, s);
-    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, s);
-    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, s);
+    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
+    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
+    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
+    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
+    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
+    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;
@@ -664,7 +666,7 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
                             void* grad_out, void* loss_count, void* margins_ws,
                             void* mult_ws, int loss_type,
                             const void* csc_colptr, const void* csc_row,
-                            const void* csc_val, void* stream) {
+                            const void* csc_val, int need_grad, void* stream) {
   (void)nnz;
   hipStream_t s = (hipStream_t)stream;
   const int* rp = (const int*)rowptr;
@@ -684,7 +686,9 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
                        (const unsigned char*)mask, loss_type, n, 1, mult,
                        (double*)loss_count);
   }
-  if (csc_colptr != nullptr) {
+  if (!need_grad) {
+    // loss-only evaluation (simple-backtracking f_x trials): skip A^T.m
+  } else if (csc_colptr != nullptr) {
     const int grid = grid_for(d, BLOCK);
     hipLaunchKernelGGL(k_csc_grad, dim3(grid), dim3(BLOCK), 0, s,
                        (const int*)csc_colptr, (const int*)csc_row,

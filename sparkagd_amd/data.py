@@ -60,8 +60,9 @@ class DenseShard:
     def nbytes(self) -> int:
         return self.features.numel() * self.features.element_size() + self.labels.numel() * self.labels.element_size()
 
-    def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None):
-        return ops.dense_eval(self.features, self.labels, w, loss_type, mask)
+    def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None,
+             need_grad: bool = True):
+        return ops.dense_eval(self.features, self.labels, w, loss_type, mask, need_grad)
 
 
 class CSRShard:
@@ -126,9 +127,11 @@ class CSRShard:
         torch.cumsum(colcounts, dim=0, out=colptr[1:])
         return colptr.to(torch.int32).contiguous(), csc_row, csc_val
 
-    def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None):
+    def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None,
+             need_grad: bool = True):
         return ops.csr_eval(self.rowptr, self.col, self.val, self.labels, w,
-                            loss_type, mask, self._d, csc=self.csc)
+                            loss_type, mask, self._d, csc=self.csc,
+                            need_grad=need_grad)
 
 
 # ---------------------------------------------------------------------------

@@ -31,15 +31,18 @@ class Gradient:
         shard,
         w: torch.Tensor,
         mask: Optional[torch.Tensor] = None,
+        need_grad: bool = True,
     ) -> Tuple[torch.Tensor, torch.Tensor]:
         """Return (grad_sum [d], loss_count float64 [2]) over the local shard.
 
         ``loss_count[0]`` is the sum of per-example losses; ``loss_count[1]``
         is the number of (unmasked) examples. Dividing by the globally
         all-reduced count happens in the optimizer, matching
-        ``AGD.scala:206-207``.
+        ``AGD.scala:206-207``. With ``need_grad=False`` only the loss side is
+        computed (one data pass instead of two — used by the simple
+        backtracking test, which needs f_x but not g_x).
         """
-        return shard.eval(w, self.LOSS_TYPE, mask)
+        return shard.eval(w, self.LOSS_TYPE, mask, need_grad)
 
     # --- MLlib per-example API parity (reference Gradient.compute) ---
     def compute(

@@ -77,10 +77,11 @@ def dense_eval(
     w: torch.Tensor,
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(features):
-        return _get_hip().dense_eval(features, labels, w, loss_type, mask)
-    return reference.dense_eval(features, labels, w, loss_type, mask)
+        return _get_hip().dense_eval(features, labels, w, loss_type, mask, need_grad)
+    return reference.dense_eval(features, labels, w, loss_type, mask, need_grad)
 
 
 def csr_eval(
@@ -93,11 +94,12 @@ def csr_eval(
     mask: Optional[torch.Tensor] = None,
     d: Optional[int] = None,
     csc=None,
+    need_grad: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(val):
-        return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, csc)
+        return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, csc, need_grad)
     # the torch reference (sparse_csr @ / .t() @) is already deterministic
-    return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d)
+    return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, need_grad)
 
 
 def prox(

@@ -11,11 +11,10 @@ Determinism notes:
   column-slab) workgroup and reduces them in a fixed order => bitwise
   reproducible gradients across runs (and therefore across ranks given
   identical inputs).
-* the CSR A^T·m path uses fp32 atomics (scatter); its summation order is
-  non-deterministic, which perturbs the gradient at the fp32-rounding level.
-  Tests compare it against the fp64 torch oracle with tolerance; a
-  deterministic CSC-transpose variant is the planned alternative
-  (SURVEY.md §5 'Race detection').
+* the CSR A^T·m path is deterministic when the shard carries its CSC copy
+  (the default: a gather kernel, no atomics). Without it an fp32 atomic
+  scatter is used, whose summation order is non-deterministic at the
+  rounding level; tests cross-check the two (SURVEY.md §5 'Race detection').
 """
 
 from __future__ import annotations
@@ -56,9 +55,9 @@ def load() -> ctypes.CDLL:
 
     P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
     lib.agd_dense_eval.restype = I
-    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, P]
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, P]
     lib.agd_csr_eval.restype = I
-    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, P]
+    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, P]
     lib.agd_axpby.restype = I
     lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
     lib.agd_prox.restype = I
@@ -101,6 +100,7 @@ def dense_eval(
     w: torch.Tensor,
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     assert features.is_cuda and features.is_contiguous() and features.ndim == 2
@@ -116,18 +116,23 @@ def dense_eval(
     mask = _prep_mask(mask, features.device)
 
     dev = features.device
-    grad = torch.empty(d, dtype=acc, device=dev)
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
     n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype))
     margins = torch.empty(n_slabs * n, dtype=acc, device=dev)
     mult = torch.empty(n, dtype=acc, device=dev)
-    n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
-    part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
+    if need_grad:
+        grad = torch.empty(d, dtype=acc, device=dev)
+        n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
+        part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
+    else:
+        grad = part = None
+        n_rb = 1
 
     rc = lib.agd_dense_eval(
         _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
-        _ptr(part), n_rb, loss_type, n_slabs, _stream(features),
+        _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
+        _stream(features),
     )
     _check(rc)
     return grad, loss_count
@@ -143,6 +148,7 @@ def csr_eval(
     mask: Optional[torch.Tensor] = None,
     d: Optional[int] = None,
     csc: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None,
+    need_grad: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     assert val.is_cuda and val.dtype == torch.float32
@@ -162,7 +168,10 @@ def csr_eval(
     mask = _prep_mask(mask, val.device)
 
     dev = val.device
-    if csc is None:
+    if not need_grad:
+        grad = None
+        cp = cr = cv = None
+    elif csc is None:
         grad = torch.zeros(d, dtype=torch.float32, device=dev)  # atomic path accumulates
         cp = cr = cv = None
     else:
@@ -176,7 +185,7 @@ def csr_eval(
         _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask),
         _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
         _ptr(margins), _ptr(mult), loss_type,
-        _ptr(cp), _ptr(cr), _ptr(cv), _stream(val),
+        _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, _stream(val),
     )
     _check(rc)
     return grad, loss_count

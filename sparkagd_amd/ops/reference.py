@@ -65,6 +65,7 @@ def dense_eval(
     w: torch.Tensor,
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Batched loss/gradient over a dense shard.
 
@@ -86,8 +87,10 @@ def dense_eval(
         count = mask.sum().to(torch.float64)
     else:
         count = torch.tensor(float(features.shape[0]), dtype=torch.float64, device=features.device)
-    grad_sum = (features.to(acc_dtype).T @ mult.to(acc_dtype)).to(w.dtype)
     loss_count = torch.stack([loss.to(torch.float64).sum(), count])
+    if not need_grad:
+        return None, loss_count
+    grad_sum = (features.to(acc_dtype).T @ mult.to(acc_dtype)).to(w.dtype)
     return grad_sum, loss_count
 
 
@@ -100,6 +103,7 @@ def csr_eval(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     d: Optional[int] = None,
+    need_grad: bool = True,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Batched loss/gradient over a CSR shard (MLlib sparse-Vector path analog)."""
     n = rowptr.numel() - 1
@@ -117,8 +121,10 @@ def csr_eval(
         count = mask.sum().to(torch.float64)
     else:
         count = torch.tensor(float(n), dtype=torch.float64, device=val.device)
-    grad_sum = (a.t() @ mult.to(acc_dtype)).to(w.dtype)
     loss_count = torch.stack([loss.to(torch.float64).sum(), count])
+    if not need_grad:
+        return None, loss_count
+    grad_sum = (a.t() @ mult.to(acc_dtype)).to(w.dtype)
     return grad_sum, loss_count
 
 

@@ -48,16 +48,23 @@ class Optimizer:
 
 def _apply_smooth(
     data, gradient: Gradient, comm: Communicator, v: torch.Tensor,
-    mask: Optional[torch.Tensor] = None,
-) -> Tuple[float, torch.Tensor, float]:
+    mask: Optional[torch.Tensor] = None, need_grad: bool = True,
+) -> Tuple[float, Optional[torch.Tensor], float]:
     """The distributed loss/gradient pass (reference ``applySmooth``,
     ``AGD.scala:192-208``): one fused kernel sequence on the local shard +
-    one all-reduce; returns (mean loss, mean gradient [device], count)."""
-    grad_sum, loss_count = gradient.eval(data, v, mask)
-    comm.allreduce_eval_(grad_sum, loss_count)
+    one all-reduce; returns (mean loss, mean gradient [device], count).
+
+    ``need_grad=False`` evaluates the loss side only (one data pass instead
+    of two): used for the simple-backtracking f_x trials, which the reference
+    pays a full gradient evaluation for (AGD.scala:269) without ever using
+    g_x in the simple test."""
+    grad_sum, loss_count = gradient.eval(data, v, mask, need_grad)
+    if grad_sum is not None:
+        comm.allreduce_(grad_sum)
+    comm.allreduce_(loss_count)
     lc = loss_count.to("cpu", non_blocking=False)  # single host sync per eval
     loss_sum, count = float(lc[0]), float(lc[1])
-    if count > 0:
+    if count > 0 and grad_sum is not None:
         grad_sum.div_(count)
     return (loss_sum / count if count > 0 else float("nan")), grad_sum, count
 
@@ -118,9 +125,9 @@ def run(
 
     eval_state = {"n": 0, "seconds": 0.0}
 
-    def apply_smooth(v, mask=None):
+    def apply_smooth(v, mask=None, need_grad=True):
         t0 = time.perf_counter()
-        out = _apply_smooth(data, gradient, comm, v, mask)
+        out = _apply_smooth(data, gradient, comm, v, mask, need_grad)
         # _apply_smooth ends with the host fetch of (loss, count), so this
         # wall segment covers the kernels + all-reduce for the evaluation.
         eval_state["n"] += 1
@@ -164,13 +171,16 @@ def run(
             if xy_sq == 0.0:
                 break
 
-            f_x, g_x, _ = apply_smooth(x)
-            f_x_bt = f_x
             if backtrack_simple:
+                # the simple test needs f_x only: loss-only pass (no A^T·m)
+                f_x, _gx_none, _ = apply_smooth(x, need_grad=False)
+                f_x_bt = f_x
                 q_x = f_y + float(scal[1]) + 0.5 * L * xy_sq
                 localL = L + 2.0 * max(f_x - q_x, 0.0) / xy_sq
                 backtrack_simple = abs(f_y - f_x) >= backtrack_tol * max(abs(f_x), abs(f_y))
             else:
+                f_x, g_x, _ = apply_smooth(x)
+                f_x_bt = f_x
                 localL = 2.0 * float(ops.dot_diff(x, y, g_x, g_y)) / xy_sq
 
             if localL <= L or L >= Lexact:

@@ -110,6 +110,17 @@ def test_csr_eval_matches_reference(loss_type):
     torch.testing.assert_close(lc_h, lc_r, rtol=1e-5, atol=1e-6)
 
 
+def test_dense_eval_loss_only():
+    """need_grad=False returns the identical loss/count without the A^T·m pass."""
+    from sparkagd_amd.ops import hiplib
+
+    A, y, w = _mk_dense(4096, 512, torch.bfloat16, seed=21)
+    gfull, lc_full = hiplib.dense_eval(A, y, w, ops.LOSS_LOGISTIC, need_grad=True)
+    gnone, lc_loss = hiplib.dense_eval(A, y, w, ops.LOSS_LOGISTIC, need_grad=False)
+    assert gnone is None and gfull is not None
+    assert torch.equal(lc_full, lc_loss)
+
+
 def test_csr_csc_deterministic_vs_atomic():
     """The CSC-gather A^T·m equals the atomic-scatter path (tolerance) and is
     bitwise reproducible run-to-run (SURVEY.md §5 race-detection cross-check)."""
