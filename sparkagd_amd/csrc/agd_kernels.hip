@@ -154,6 +154,33 @@ __device__ __forceinline__ void loadAcc(const TACC* __restrict__ p, TACC (&out)[
   }
 }
 
+// Vector store, mirror of loadAcc.
+template <typename TACC, int W>
+__device__ __forceinline__ void storeAcc(TACC* __restrict__ p, const TACC (&v)[W]) {
+  constexpr int EPC = 16 / sizeof(TACC);
+  if constexpr (W >= EPC) {
+#pragma unroll
+    for (int ch = 0; ch < W / EPC; ++ch) {
+      if constexpr (sizeof(TACC) == 4) {
+        using f32x4 = __attribute__((ext_vector_type(4))) float;
+        f32x4 o;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) o[k] = v[ch * 4 + k];
+        *(f32x4*)(p + ch * 4) = o;
+      } else {
+        using f64x2 = __attribute__((ext_vector_type(2))) double;
+        f64x2 o;
+#pragma unroll
+        for (int k = 0; k < 2; ++k) o[k] = v[ch * 2 + k];
+        *(f64x2*)(p + ch * 2) = o;
+      }
+    }
+  } else {
+#pragma unroll
+    for (int k = 0; k < W; ++k) p[k] = v[k];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // K1a: dense margins  z[r] = <A[r,:], w>
 //
@@ -437,10 +464,21 @@ template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_axpby(double a, const T* __restrict__ x,
                                                  double b, const T* __restrict__ y,
                                                  T* __restrict__ out, ll n) {
+  constexpr int VE = 16 / sizeof(T);
   const ll stride = (ll)gridDim.x * BLOCK;
+  const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
   const T ta = (T)a, tb = (T)b;
-  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride)
-    out[i] = ta * x[i] + tb * y[i];
+  const ll nv = n / VE;
+  for (ll i = gid; i < nv; i += stride) {
+    T xv[VE], yv[VE], ov[VE];
+    loadAcc<T, VE>(x + i * VE, xv);
+    loadAcc<T, VE>(y + i * VE, yv);
+#pragma unroll
+    for (int k = 0; k < VE; ++k) ov[k] = ta * xv[k] + tb * yv[k];
+    storeAcc<T, VE>(out + i * VE, ov);
+  }
+  const ll t = nv * VE + gid;  // tail (< VE elements)
+  if (t < n) out[t] = ta * x[t] + tb * y[t];
 }
 
 // ---------------------------------------------------------------------------
@@ -454,30 +492,43 @@ __global__ __launch_bounds__(BLOCK) void k_axpby(double a, const T* __restrict__
 #define PROX_L2 2
 
 template <typename T>
+__device__ __forceinline__ T prox_elem(int kind, T wi, T gi, T ts, T tl,
+                                       double& racc) {
+  if (kind == PROX_SIMPLE) return wi - ts * gi;
+  if (kind == PROX_L1) {
+    const T w1 = wi - ts * gi;
+    const T aw = fabs(w1) - tl * ts;
+    const T wn = (aw > (T)0) ? ((w1 > (T)0) ? aw : -aw) : (T)0;
+    racc += (double)tl * fabs((double)wn);
+    return wn;
+  }
+  const T wn = wi * ((T)1 - ts * tl) - ts * gi;  // PROX_L2
+  racc += 0.5 * (double)tl * (double)wn * (double)wn;
+  return wn;
+}
+
+template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_prox(int kind, const T* __restrict__ w,
                                                 const T* __restrict__ g,
                                                 double step, double lam,
                                                 T* __restrict__ out,
                                                 double* __restrict__ reg, ll n) {
+  constexpr int VE = 16 / sizeof(T);
   const ll stride = (ll)gridDim.x * BLOCK;
+  const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
   const T ts = (T)step, tl = (T)lam;
   double racc = 0.0;
-  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
-    if (kind == PROX_SIMPLE) {
-      out[i] = w[i] - ts * g[i];
-    } else if (kind == PROX_L1) {
-      const T w1 = w[i] - ts * g[i];
-      const T shrink = tl * ts;
-      const T aw = fabs(w1) - shrink;
-      const T wn = (aw > (T)0) ? ((w1 > (T)0) ? aw : -aw) : (T)0;
-      out[i] = wn;
-      racc += (double)tl * fabs((double)wn);
-    } else {  // PROX_L2
-      const T wn = w[i] * ((T)1 - ts * tl) - ts * g[i];
-      out[i] = wn;
-      racc += 0.5 * (double)tl * (double)wn * (double)wn;
-    }
+  const ll nv = n / VE;
+  for (ll i = gid; i < nv; i += stride) {
+    T wv[VE], gv[VE], ov[VE];
+    loadAcc<T, VE>(w + i * VE, wv);
+    loadAcc<T, VE>(g + i * VE, gv);
+#pragma unroll
+    for (int k = 0; k < VE; ++k) ov[k] = prox_elem(kind, wv[k], gv[k], ts, tl, racc);
+    storeAcc<T, VE>(out + i * VE, ov);
   }
+  const ll t = nv * VE + gid;
+  if (t < n) out[t] = prox_elem(kind, w[t], g[t], ts, tl, racc);
   if (kind != PROX_SIMPLE) {
     double acc[1] = {racc};
     block_reduce_atomic<1>(acc, reg);
@@ -493,22 +544,38 @@ __global__ __launch_bounds__(BLOCK) void k_prox(int kind, const T* __restrict__ 
 // ---------------------------------------------------------------------------
 
 template <typename T>
+__device__ __forceinline__ void fused_scalars_elem(double xi, double yi,
+                                                   double gi, double oi,
+                                                   double (&a)[5]) {
+  const double xy = xi - yi, dx = xi - oi;
+  a[0] += xy * xy;
+  a[1] += xy * gi;
+  a[2] += xi * xi;
+  a[3] += dx * dx;
+  a[4] += gi * dx;
+}
+
+template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_fused_scalars(
     const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ gy,
     const T* __restrict__ xold, double* __restrict__ out, ll n) {
-  double a0 = 0, a1 = 0, a2 = 0, a3 = 0, a4 = 0;
+  constexpr int VE = 16 / sizeof(T);
+  double acc[5] = {0, 0, 0, 0, 0};
   const ll stride = (ll)gridDim.x * BLOCK;
-  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
-    const double xi = (double)x[i], yi = (double)y[i];
-    const double gi = (double)gy[i], oi = (double)xold[i];
-    const double xy = xi - yi, dx = xi - oi;
-    a0 += xy * xy;
-    a1 += xy * gi;
-    a2 += xi * xi;
-    a3 += dx * dx;
-    a4 += gi * dx;
+  const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
+  const ll nv = n / VE;
+  for (ll i = gid; i < nv; i += stride) {
+    T xv[VE], yv[VE], gv[VE], ov[VE];
+    loadAcc<T, VE>(x + i * VE, xv);
+    loadAcc<T, VE>(y + i * VE, yv);
+    loadAcc<T, VE>(gy + i * VE, gv);
+    loadAcc<T, VE>(xold + i * VE, ov);
+#pragma unroll
+    for (int k = 0; k < VE; ++k)
+      fused_scalars_elem<T>(xv[k], yv[k], gv[k], ov[k], acc);
   }
-  double acc[5] = {a0, a1, a2, a3, a4};
+  const ll t = nv * VE + gid;
+  if (t < n) fused_scalars_elem<T>(x[t], y[t], gy[t], xold[t], acc);
   block_reduce_atomic<5>(acc, out);
 }
 
@@ -516,10 +583,24 @@ template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_dot_diff(
     const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ gx,
     const T* __restrict__ gy, double* __restrict__ out, ll n) {
+  constexpr int VE = 16 / sizeof(T);
   double a = 0;
   const ll stride = (ll)gridDim.x * BLOCK;
-  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride)
-    a += ((double)x[i] - (double)y[i]) * ((double)gx[i] - (double)gy[i]);
+  const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
+  const ll nv = n / VE;
+  for (ll i = gid; i < nv; i += stride) {
+    T xv[VE], yv[VE], gxv[VE], gyv[VE];
+    loadAcc<T, VE>(x + i * VE, xv);
+    loadAcc<T, VE>(y + i * VE, yv);
+    loadAcc<T, VE>(gx + i * VE, gxv);
+    loadAcc<T, VE>(gy + i * VE, gyv);
+#pragma unroll
+    for (int k = 0; k < VE; ++k)
+      a += ((double)xv[k] - (double)yv[k]) * ((double)gxv[k] - (double)gyv[k]);
+  }
+  const ll t = nv * VE + gid;
+  if (t < n)
+    a += ((double)x[t] - (double)y[t]) * ((double)gx[t] - (double)gy[t]);
   double acc[1] = {a};
   block_reduce_atomic<1>(acc, out);
 }
