@@ -265,6 +265,76 @@ __global__ __launch_bounds__(BLOCK) void k_dense_margins(
 }
 
 // ---------------------------------------------------------------------------
+// K1a-MFMA: dense margins on the matrix cores (bf16 shards).
+//
+// D[32x32] = A_frag[32x16] · B_frag[16x32] with v_mfma_f32_32x32x16_bf16:
+// each wave owns a 32-row group; the A fragment IS the natural coalesced
+// 16-B-per-lane stream of the shard (lane l reads A[r0 + (l&31)][k + 8*(l>>5)
+// .. +7]); the weight vector is staged through LDS in 16 KiB bf16 tiles
+// (converted once per block, broadcast-read by all 4 waves) and replicated
+// across the 32 B-columns, so column 0 of the accumulator is the margin.
+// Fragment layouts verified empirically by benchmarks/mfma_probe.hip (no ISA
+// doc in this environment). Requires d % 16 == 0; bf16 only. Numerics note:
+// the MFMA path rounds w to bf16 (hardware input format); the VALU path keeps
+// w fp32 — both pass the parity tests, and both run at the HBM roofline
+// (arithmetic intensity ~1 FLOP/byte), so the selector is a measurement
+// question, not a throughput one.
+// ---------------------------------------------------------------------------
+
+#define MFMA_KCHUNK 8192  // w bf16 elements staged per LDS tile (16 KiB)
+#define MFMA_ROWS 128     // rows per 4-wave block (32 per wave)
+
+__device__ __forceinline__ ubf16 f2bf_rne(float f) {
+  union { float f; unsigned u; } v{f};
+  return (ubf16)((v.u + 0x7FFF + ((v.u >> 16) & 1)) >> 16);
+}
+
+using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+using f32x16_t = __attribute__((ext_vector_type(16))) float;
+
+__global__ __launch_bounds__(BLOCK) void k_dense_margins_mfma(
+    const ubf16* __restrict__ A, const float* __restrict__ w, ll n, ll d,
+    ll slab_w, int n_slabs, float* __restrict__ part) {
+  __shared__ ubf16 wlds[MFMA_KCHUNK];
+  const int l = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const ll n_rb = (n + MFMA_ROWS - 1) / MFMA_ROWS;
+  const ll n_tasks = n_rb * n_slabs;
+  for (ll t = blockIdx.x; t < n_tasks; t += gridDim.x) {
+    const ll rb = t / n_slabs;
+    const ll s = t - rb * n_slabs;
+    const ll c_lo = s * slab_w;
+    const ll c_hi = (c_lo + slab_w < d) ? c_lo + slab_w : d;
+    const ll r0 = rb * MFMA_ROWS + (ll)wid * 32;
+    // lane's row (clamped on the ragged tail block: its garbage products land
+    // in accumulator entries whose rows are never stored)
+    const ll row = (r0 + (l & 31) < n) ? r0 + (l & 31) : n - 1;
+    const ubf16* __restrict__ arow = A + row * d + 8 * (l >> 5);
+    f32x16_t acc = {};
+    for (ll kc = c_lo; kc < c_hi; kc += MFMA_KCHUNK) {
+      const ll kcn = (kc + MFMA_KCHUNK <= c_hi) ? MFMA_KCHUNK : (c_hi - kc);
+      __syncthreads();
+      for (ll k = threadIdx.x; k < kcn; k += BLOCK) wlds[k] = f2bf_rne(w[kc + k]);
+      __syncthreads();
+      const int bbase = 8 * (l >> 5);
+      for (ll kk = 0; kk + 16 <= kcn; kk += 16) {
+        bf16x8_t a = *(const bf16x8_t*)(arow + kc + kk);
+        bf16x8_t b = *(const bf16x8_t*)&wlds[kk + bbase];
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+      }
+    }
+    // column 0 of D holds the margins; lanes 0 and 32 carry all 32 rows.
+    if ((l & 31) == 0) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const ll rg = r0 + (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+        if (rg < n) part[s * n + rg] = acc[r];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // K1b: elementwise multiplier + loss/count reduction
 //
 // loss conventions (z = <w, x_i>; identical algebra to MLlib 1.3's
@@ -641,17 +711,36 @@ extern "C" long long agd_dense_rowblocks(long long n, long long d, int a_dtype) 
   return n_rb;
 }
 
+// margins_algo: 0 = auto, 1 = VALU row-group kernel, 2 = MFMA kernel.
+// MFMA eligibility: bf16 shard with d % 16 == 0.
+static inline int margins_algo_eff(int algo, int a_dtype, ll d) {
+  const bool mfma_ok = (a_dtype == 0) && (d % 16 == 0);
+  if (algo == 2 && !mfma_ok) return 1;
+  if (algo == 0) return mfma_ok ? 2 : 1;  // auto: MFMA (ties VALU at the
+                                          // HBM roofline; measured round 1)
+  return algo;
+}
+
 // Margin-pass column slabs (>1 only for thin-n / fat-d shards, where
 // row-group parallelism alone cannot fill 256 CUs). Callers size the margins
 // workspace as n_slabs * n accumulators.
-extern "C" int agd_margin_slabs(long long n, long long d, int a_dtype) {
-  const ll target_waves = 16384;
-  const ll n_rg = (n + MARGIN_ROWS - 1) / MARGIN_ROWS;
-  if (n_rg >= target_waves) return 1;
-  const int w = pick_w(a_dtype, d);
-  const ll min_slab = (ll)WAVE * w * 4;  // keep >= 4 vector iterations per slab
+extern "C" int agd_margin_slabs(long long n, long long d, int a_dtype,
+                                int margins_algo) {
+  const int algo = margins_algo_eff(margins_algo, a_dtype, (ll)d);
+  ll n_units, target, min_slab;
+  if (algo == 2) {
+    n_units = (n + MFMA_ROWS - 1) / MFMA_ROWS;  // 4-wave blocks
+    target = 2048;                               // ~8 blocks/CU
+    min_slab = 4096;
+  } else {
+    n_units = (n + MARGIN_ROWS - 1) / MARGIN_ROWS;  // waves
+    target = 16384;
+    const int w = pick_w(a_dtype, d);
+    min_slab = (ll)WAVE * w * 4;
+  }
+  if (n_units >= target) return 1;
   ll max_slabs = (d + min_slab - 1) / min_slab;
-  ll want = (target_waves + n_rg - 1) / n_rg;
+  ll want = (target + n_units - 1) / n_units;
   ll s = want < max_slabs ? want : max_slabs;
   if (s < 1) s = 1;
   if (s > 1024) s = 1024;
@@ -663,13 +752,16 @@ static int dense_eval_t(const void* A, const float* labels,
                         const unsigned char* mask, const void* w, ll n, ll d,
                         void* grad_out, double* loss_count, void* margins_ws,
                         void* mult_ws, void* part_ws, ll n_rb, int loss_type,
-                        int n_slabs, int need_grad, hipStream_t stream) {
+                        int n_slabs, int need_grad, int margins_algo,
+                        hipStream_t stream) {
   const TA* a = (const TA*)A;
   const TACC* wp = (const TACC*)w;
   TACC* margins = (TACC*)margins_ws;
   TACC* mult = (TACC*)mult_ws;
   TACC* grad = (TACC*)grad_out;
   TACC* part = (n_rb == 1) ? grad : (TACC*)part_ws;
+  const int algo = margins_algo_eff(margins_algo, sizeof(TA) == 2 ? 0 : (sizeof(TA) == 4 ? 1 : 2), d);
+  const bool use_mfma = (algo == 2) && (sizeof(TA) == 2) && (W == 8);
 
   ll slab_w = d;
   if (n_slabs > 1) {
@@ -679,7 +771,13 @@ static int dense_eval_t(const void* A, const float* labels,
     n_slabs = (int)((d + slab_w - 1) / slab_w);
   }
 
-  {
+  if (use_mfma) {
+    const ll tasks = ((n + MFMA_ROWS - 1) / MFMA_ROWS) * n_slabs;
+    const int grid = grid_for(tasks, 1);
+    hipLaunchKernelGGL(k_dense_margins_mfma, dim3(grid), dim3(BLOCK), 0,
+                       stream, (const ubf16*)A, (const float*)w, n, d, slab_w,
+                       n_slabs, (float*)margins);
+  } else {
     const ll tasks = ((n + MARGIN_ROWS - 1) / MARGIN_ROWS) * n_slabs;
     const int grid = grid_for(tasks, WAVES_PER_BLOCK);
     hipLaunchKernelGGL((k_dense_margins<TA, TACC, W>), dim3(grid), dim3(BLOCK),
@@ -717,19 +815,19 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
                               long long d, void* grad_out, void* loss_count,
                               void* margins_ws, void* mult_ws, void* part_ws,
                               long long n_rb, int loss_type, int n_slabs,
-                              int need_grad, void* stream) {
+                              int need_grad, int margins_algo, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const float* lab = (const float*)labels;
   const unsigned char* msk = (const unsigned char*)mask;
   double* lc = (double*)loss_count;
   const int W = pick_w(a_dtype, d);
   switch (a_dtype * 10 + (W > 1 ? 1 : 0)) {
-    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
-    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
-    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
-    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
-    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
-    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, s);
+    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
+    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
+    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
+    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
+    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
+    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;

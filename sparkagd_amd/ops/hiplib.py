@@ -51,11 +51,11 @@ def load() -> ctypes.CDLL:
     lib.agd_dense_rowblocks.restype = ctypes.c_longlong
     lib.agd_dense_rowblocks.argtypes = [ctypes.c_longlong, ctypes.c_longlong, ctypes.c_int]
     lib.agd_margin_slabs.restype = ctypes.c_int
-    lib.agd_margin_slabs.argtypes = [ctypes.c_longlong, ctypes.c_longlong, ctypes.c_int]
+    lib.agd_margin_slabs.argtypes = [ctypes.c_longlong, ctypes.c_longlong, ctypes.c_int, ctypes.c_int]
 
     P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
     lib.agd_dense_eval.restype = I
-    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, P]
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, P]
     lib.agd_csr_eval.restype = I
     lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, P]
     lib.agd_axpby.restype = I
@@ -117,7 +117,9 @@ def dense_eval(
 
     dev = features.device
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
-    n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype))
+    # margins algorithm: 0 auto, 1 VALU row-group, 2 MFMA (bf16, d%16==0)
+    margins_algo = int(os.environ.get("SPARKAGD_MARGINS_ALGO", "0"))
+    n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype, margins_algo))
     margins = torch.empty(n_slabs * n, dtype=acc, device=dev)
     mult = torch.empty(n, dtype=acc, device=dev)
     if need_grad:
@@ -132,7 +134,7 @@ def dense_eval(
         _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
         _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
-        _stream(features),
+        margins_algo, _stream(features),
     )
     _check(rc)
     return grad, loss_count

@@ -110,6 +110,28 @@ def test_csr_eval_matches_reference(loss_type):
     torch.testing.assert_close(lc_h, lc_r, rtol=1e-5, atol=1e-6)
 
 
+@pytest.mark.parametrize("n,d", [(8192, 1024), (100, 20000)])
+def test_margins_mfma_matches_valu(n, d, monkeypatch):
+    """The MFMA margins kernel (32x32x16 bf16, LDS-staged w tiles) agrees with
+    the VALU kernel up to w's bf16 rounding (the MFMA input format)."""
+    from sparkagd_amd.ops import hiplib, reference
+
+    A, y, w = _mk_dense(n, d, torch.bfloat16, seed=31)
+    monkeypatch.setenv("SPARKAGD_MARGINS_ALGO", "1")
+    g1, lc1 = hiplib.dense_eval(A, y, w, ops.LOSS_LOGISTIC)
+    monkeypatch.setenv("SPARKAGD_MARGINS_ALGO", "2")
+    g2, lc2 = hiplib.dense_eval(A, y, w, ops.LOSS_LOGISTIC)
+    monkeypatch.delenv("SPARKAGD_MARGINS_ALGO")
+    # oracle with w pre-rounded to bf16 (what the matrix core consumes)
+    w_rounded = w.to(torch.bfloat16).to(torch.float32)
+    gr, lcr = reference.dense_eval(A, y, w_rounded, ops.LOSS_LOGISTIC)
+    torch.testing.assert_close(g2, gr, rtol=3e-4, atol=3e-3)
+    torch.testing.assert_close(lc2, lcr, rtol=1e-5, atol=1e-5)
+    # and the two kernels agree within the w-rounding perturbation
+    torch.testing.assert_close(g2, g1, rtol=5e-3, atol=5e-2)
+    assert float(lc1[1]) == float(lc2[1]) == n
+
+
 def test_dense_eval_loss_only():
     """need_grad=False returns the identical loss/count without the A^T·m pass."""
     from sparkagd_amd.ops import hiplib
