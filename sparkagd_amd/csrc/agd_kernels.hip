@@ -716,8 +716,12 @@ extern "C" long long agd_dense_rowblocks(long long n, long long d, int a_dtype) 
 static inline int margins_algo_eff(int algo, int a_dtype, ll d) {
   const bool mfma_ok = (a_dtype == 0) && (d % 16 == 0);
   if (algo == 2 && !mfma_ok) return 1;
-  if (algo == 0) return mfma_ok ? 2 : 1;  // auto: MFMA (ties VALU at the
-                                          // HBM roofline; measured round 1)
+  // auto = VALU: both paths are HBM-bound, and measured round 1 the VALU
+  // row-group kernel streams at 97% of the copy ceiling while the MFMA
+  // variant pays for its LDS w-staging and lower row-parallelism (~2.2x on
+  // the margins pass at d=1e6) AND rounds w to bf16 (matrix-core input
+  // format). The MFMA kernel remains selectable (algo=2) and tested.
+  if (algo == 0) return 1;
   return algo;
 }
 
