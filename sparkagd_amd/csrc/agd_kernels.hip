@@ -105,24 +105,24 @@ __device__ __forceinline__ void block_reduce_atomic(double (&acc)[NACC],
 
 // Vector load of W elements of TA starting at p (16-byte pattern for W>1),
 // converted to the accumulator type TACC.
-template <typename TA, typename TACC, int W>
+template <typename TA, typename TACC, int W, bool NT = false>
 __device__ __forceinline__ void loadW(const TA* __restrict__ p, TACC (&out)[W]) {
   if constexpr (W == 1) {
     if constexpr (sizeof(TA) == 2) out[0] = bf2f(*(const ubf16*)p);
     else out[0] = (TACC)p[0];
   } else if constexpr (sizeof(TA) == 2) {  // bf16, W == 8 (16 B)
     using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
-    u16x8 v = *(const u16x8*)p;
+    u16x8 v = NT ? __builtin_nontemporal_load((const u16x8*)p) : *(const u16x8*)p;
 #pragma unroll
     for (int k = 0; k < 8; ++k) out[k] = bf2f((ubf16)v[k]);
   } else if constexpr (sizeof(TA) == 4) {  // f32, W == 4 (16 B)
     using f32x4 = __attribute__((ext_vector_type(4))) float;
-    f32x4 v = *(const f32x4*)p;
+    f32x4 v = NT ? __builtin_nontemporal_load((const f32x4*)p) : *(const f32x4*)p;
 #pragma unroll
     for (int k = 0; k < 4; ++k) out[k] = (TACC)v[k];
   } else {  // f64, W == 2 (16 B)
     using f64x2 = __attribute__((ext_vector_type(2))) double;
-    f64x2 v = *(const f64x2*)p;
+    f64x2 v = NT ? __builtin_nontemporal_load((const f64x2*)p) : *(const f64x2*)p;
 #pragma unroll
     for (int k = 0; k < 2; ++k) out[k] = (TACC)v[k];
   }
@@ -194,7 +194,7 @@ __device__ __forceinline__ void storeAcc(TACC* __restrict__ p, const TACC (&v)[W
 
 #define MARGIN_ROWS 4  // R: rows per wave (w-load amortization)
 
-template <typename TA, typename TACC, int W>
+template <typename TA, typename TACC, int W, bool NT>
 __global__ __launch_bounds__(BLOCK) void k_dense_margins(
     const TA* __restrict__ A, const TACC* __restrict__ w, ll n, ll d,
     ll slab_w, int n_slabs, TACC* __restrict__ part) {
@@ -224,7 +224,7 @@ __global__ __launch_bounds__(BLOCK) void k_dense_margins(
 #pragma unroll
         for (int j = 0; j < R; ++j) {
           TACC v[W];
-          loadW<TA, TACC, W>(row0 + (ll)j * d + c, v);
+          loadW<TA, TACC, W, NT>(row0 + (ll)j * d + c, v);
 #pragma unroll
           for (int k = 0; k < W; ++k) acc[j] += v[k] * wv[k];
         }
@@ -408,7 +408,7 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
 // HBM traffic. A second kernel reduces over row blocks => deterministic.
 // ---------------------------------------------------------------------------
 
-template <typename TA, typename TACC, int W>
+template <typename TA, typename TACC, int W, bool NT>
 __global__ __launch_bounds__(BLOCK) void k_dense_grad(
     const TA* __restrict__ A, const TACC* __restrict__ mult, ll n, ll d,
     ll n_rb, TACC* __restrict__ part) {
@@ -430,7 +430,7 @@ __global__ __launch_bounds__(BLOCK) void k_dense_grad(
       const TA* __restrict__ p = A + r * d + c0;
       if (full) {
         TACC v[W];
-        loadW<TA, TACC, W>(p, v);
+        loadW<TA, TACC, W, NT>(p, v);
 #pragma unroll
         for (int k = 0; k < W; ++k) acc[k] += m * v[k];
       } else {
@@ -757,7 +757,7 @@ static int dense_eval_t(const void* A, const float* labels,
                         void* grad_out, double* loss_count, void* margins_ws,
                         void* mult_ws, void* part_ws, ll n_rb, int loss_type,
                         int n_slabs, int need_grad, int margins_algo,
-                        hipStream_t stream) {
+                        int nt_loads, hipStream_t stream) {
   const TA* a = (const TA*)A;
   const TACC* wp = (const TACC*)w;
   TACC* margins = (TACC*)margins_ws;
@@ -784,8 +784,14 @@ static int dense_eval_t(const void* A, const float* labels,
   } else {
     const ll tasks = ((n + MARGIN_ROWS - 1) / MARGIN_ROWS) * n_slabs;
     const int grid = grid_for(tasks, WAVES_PER_BLOCK);
-    hipLaunchKernelGGL((k_dense_margins<TA, TACC, W>), dim3(grid), dim3(BLOCK),
-                       0, stream, a, wp, n, d, slab_w, n_slabs, margins);
+    if (nt_loads)
+      hipLaunchKernelGGL((k_dense_margins<TA, TACC, W, true>), dim3(grid),
+                         dim3(BLOCK), 0, stream, a, wp, n, d, slab_w, n_slabs,
+                         margins);
+    else
+      hipLaunchKernelGGL((k_dense_margins<TA, TACC, W, false>), dim3(grid),
+                         dim3(BLOCK), 0, stream, a, wp, n, d, slab_w, n_slabs,
+                         margins);
   }
   {
     const int grid = grid_for(n, BLOCK);
@@ -798,8 +804,12 @@ static int dense_eval_t(const void* A, const float* labels,
       const ll cols_per_block = (ll)BLOCK * W;
       const ll n_cs = (d + cols_per_block - 1) / cols_per_block;
       const int grid = grid_for(n_rb * n_cs, 1);
-      hipLaunchKernelGGL((k_dense_grad<TA, TACC, W>), dim3(grid), dim3(BLOCK),
-                         0, stream, a, mult, n, d, n_rb, part);
+      if (nt_loads)
+        hipLaunchKernelGGL((k_dense_grad<TA, TACC, W, true>), dim3(grid),
+                           dim3(BLOCK), 0, stream, a, mult, n, d, n_rb, part);
+      else
+        hipLaunchKernelGGL((k_dense_grad<TA, TACC, W, false>), dim3(grid),
+                           dim3(BLOCK), 0, stream, a, mult, n, d, n_rb, part);
     }
     if (n_rb > 1) {
       const int grid = grid_for(d, BLOCK);
@@ -819,19 +829,20 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
                               long long d, void* grad_out, void* loss_count,
                               void* margins_ws, void* mult_ws, void* part_ws,
                               long long n_rb, int loss_type, int n_slabs,
-                              int need_grad, int margins_algo, void* stream) {
+                              int need_grad, int margins_algo, int nt_loads,
+                              void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const float* lab = (const float*)labels;
   const unsigned char* msk = (const unsigned char*)mask;
   double* lc = (double*)loss_count;
   const int W = pick_w(a_dtype, d);
   switch (a_dtype * 10 + (W > 1 ? 1 : 0)) {
-    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
-    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
-    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
-    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
-    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
-    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, s);
+    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
+    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
+    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
+    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
+    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
+    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;

@@ -55,7 +55,7 @@ def load() -> ctypes.CDLL:
 
     P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
     lib.agd_dense_eval.restype = I
-    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, P]
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, P]
     lib.agd_csr_eval.restype = I
     lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, P]
     lib.agd_axpby.restype = I
@@ -134,7 +134,8 @@ def dense_eval(
         _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
         _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
-        margins_algo, _stream(features),
+        margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "0")),
+        _stream(features),
     )
     _check(rc)
     return grad, loss_count
