@@ -134,7 +134,7 @@ def dense_eval(
         _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
         _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
-        margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "0")),
+        margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),  # nt A-stream: +6-9% measured
         _stream(features),
     )
     _check(rc)
