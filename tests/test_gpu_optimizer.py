@@ -75,6 +75,17 @@ def test_minibatch_sampling_on_gpu():
     assert len(hist) == 10 and hist[-1] < hist[0]
 
 
+def test_agd_on_csr_shard_gpu():
+    from sparkagd_amd.data import generate_csr_problem
+    from sparkagd_amd import ops
+
+    shard, _ = generate_csr_problem(50000, 20000, 16, seed=9, device=DEV)
+    w0 = torch.zeros(20000, device=DEV, dtype=torch.float32)
+    w, hist = run(shard, LogisticGradient(), SimpleUpdater(), 1e-10, 10, 0.0,
+                  w0, 1.0, math.inf, 0.5, 0.9, True)
+    assert hist[-1] < hist[0] and hist[-1] < math.log(2.0)
+
+
 def test_checkpoint_roundtrip_gpu(tmp_path):
     shard, _ = generate_dense_problem(n=20000, d=128, seed=4, device=DEV,
                                       dtype=torch.float32)
