@@ -65,6 +65,15 @@ class CountingGradient:
         self.n_passes += 2 if need_grad else 1
         return self.inner.eval(shard, w, mask, need_grad)
 
+    def margins(self, shard, v):
+        self.n_passes += 1
+        return self.inner.margins(shard, v)
+
+    def eval_from_margins(self, shard, margins, mask=None, need_grad=True):
+        self.n_evals += 1
+        self.n_passes += 1 if need_grad else 0
+        return self.inner.eval_from_margins(shard, margins, mask, need_grad)
+
 
 def sync(device: torch.device) -> None:
     if device.type == "cuda":

@@ -334,6 +334,19 @@ __global__ __launch_bounds__(BLOCK) void k_dense_margins_mfma(
   }
 }
 
+// Final margins from per-slab partials (margins-only mode): in-place onto
+// the s=0 region of part.
+template <typename TACC>
+__global__ __launch_bounds__(BLOCK) void k_slab_reduce(TACC* __restrict__ part,
+                                                       ll n, int n_slabs) {
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll r = (ll)blockIdx.x * BLOCK + threadIdx.x; r < n; r += stride) {
+    TACC v = part[r];
+    for (int ss = 1; ss < n_slabs; ++ss) v += part[(ll)ss * n + r];
+    part[r] = v;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // K1b: elementwise multiplier + loss/count reduction
 //
@@ -757,7 +770,7 @@ static int dense_eval_t(const void* A, const float* labels,
                         void* grad_out, double* loss_count, void* margins_ws,
                         void* mult_ws, void* part_ws, ll n_rb, int loss_type,
                         int n_slabs, int need_grad, int margins_algo,
-                        int nt_loads, hipStream_t stream) {
+                        int nt_loads, int mode, hipStream_t stream) {
   const TA* a = (const TA*)A;
   const TACC* wp = (const TACC*)w;
   TACC* margins = (TACC*)margins_ws;
@@ -768,6 +781,7 @@ static int dense_eval_t(const void* A, const float* labels,
   const bool use_mfma = (algo == 2) && (sizeof(TA) == 2) && (W == 8);
 
   ll slab_w = d;
+  if (mode == 2) n_slabs = 1;  // margins provided by the caller
   if (n_slabs > 1) {
     slab_w = (d + n_slabs - 1) / n_slabs;
     const ll align = (ll)WAVE * W;
@@ -775,7 +789,9 @@ static int dense_eval_t(const void* A, const float* labels,
     n_slabs = (int)((d + slab_w - 1) / slab_w);
   }
 
-  if (use_mfma) {
+  if (mode == 2) {
+    // skip the margins pass entirely
+  } else if (use_mfma) {
     const ll tasks = ((n + MFMA_ROWS - 1) / MFMA_ROWS) * n_slabs;
     const int grid = grid_for(tasks, 1);
     hipLaunchKernelGGL(k_dense_margins_mfma, dim3(grid), dim3(BLOCK), 0,
@@ -792,6 +808,15 @@ static int dense_eval_t(const void* A, const float* labels,
       hipLaunchKernelGGL((k_dense_margins<TA, TACC, W, false>), dim3(grid),
                          dim3(BLOCK), 0, stream, a, wp, n, d, slab_w, n_slabs,
                          margins);
+  }
+  if (mode == 1) {  // margins only: reduce slabs in place and return
+    if (n_slabs > 1) {
+      const int grid = grid_for(n, BLOCK);
+      hipLaunchKernelGGL((k_slab_reduce<TACC>), dim3(grid), dim3(BLOCK), 0,
+                         stream, margins, n, n_slabs);
+    }
+    HIP_CHECK(hipGetLastError());
+    return 0;
   }
   {
     const int grid = grid_for(n, BLOCK);
@@ -830,19 +855,19 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
                               void* margins_ws, void* mult_ws, void* part_ws,
                               long long n_rb, int loss_type, int n_slabs,
                               int need_grad, int margins_algo, int nt_loads,
-                              void* stream) {
+                              int mode, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const float* lab = (const float*)labels;
   const unsigned char* msk = (const unsigned char*)mask;
   double* lc = (double*)loss_count;
   const int W = pick_w(a_dtype, d);
   switch (a_dtype * 10 + (W > 1 ? 1 : 0)) {
-    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
-    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
-    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
-    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
-    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
-    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, s);
+    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
+    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
+    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
+    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
+    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
+    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;
@@ -860,7 +885,8 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
                             void* grad_out, void* loss_count, void* margins_ws,
                             void* mult_ws, int loss_type,
                             const void* csc_colptr, const void* csc_row,
-                            const void* csc_val, int need_grad, void* stream) {
+                            const void* csc_val, int need_grad, int mode,
+                            void* stream) {
   (void)nnz;
   hipStream_t s = (hipStream_t)stream;
   const int* rp = (const int*)rowptr;
@@ -868,10 +894,14 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
   const float* v = (const float*)val;
   float* margins = (float*)margins_ws;
   float* mult = (float*)mult_ws;
-  {
+  if (mode != 2) {
     const int grid = grid_for(n, WAVES_PER_BLOCK);
     hipLaunchKernelGGL(k_csr_margins, dim3(grid), dim3(BLOCK), 0, s, rp, ci, v,
                        (const float*)w, n, margins);
+  }
+  if (mode == 1) {
+    HIP_CHECK(hipGetLastError());
+    return 0;
   }
   {
     const int grid = grid_for(n, BLOCK);

@@ -64,6 +64,16 @@ class DenseShard:
              need_grad: bool = True):
         return ops.dense_eval(self.features, self.labels, w, loss_type, mask, need_grad)
 
+    # --- margin-state tracking support (one data pass instead of two when
+    # the caller already holds A @ w; see optimizer.py 'track_margins') ---
+    def margins(self, v: torch.Tensor) -> torch.Tensor:
+        return ops.dense_margins(self.features, v)
+
+    def eval_from_margins(self, margins: torch.Tensor, loss_type: int,
+                          mask: Optional[torch.Tensor] = None, need_grad: bool = True):
+        return ops.dense_eval_from_margins(self.features, margins, self.labels,
+                                           loss_type, mask, need_grad)
+
 
 class CSRShard:
     """One rank's rows of a CSR sparse design matrix.
@@ -132,6 +142,15 @@ class CSRShard:
         return ops.csr_eval(self.rowptr, self.col, self.val, self.labels, w,
                             loss_type, mask, self._d, csc=self.csc,
                             need_grad=need_grad)
+
+    def margins(self, v: torch.Tensor) -> torch.Tensor:
+        return ops.csr_margins(self.rowptr, self.col, self.val, v)
+
+    def eval_from_margins(self, margins: torch.Tensor, loss_type: int,
+                          mask: Optional[torch.Tensor] = None, need_grad: bool = True):
+        return ops.csr_eval_from_margins(self.rowptr, self.col, self.val, margins,
+                                         self.labels, loss_type, mask, self._d,
+                                         csc=self.csc, need_grad=need_grad)
 
 
 # ---------------------------------------------------------------------------

@@ -44,6 +44,16 @@ class Gradient:
         """
         return shard.eval(w, self.LOSS_TYPE, mask, need_grad)
 
+    def margins(self, shard, v: torch.Tensor) -> torch.Tensor:
+        """A @ v over the local shard (margin-state tracking)."""
+        return shard.margins(v)
+
+    def eval_from_margins(self, shard, margins: torch.Tensor,
+                          mask: Optional[torch.Tensor] = None,
+                          need_grad: bool = True):
+        """Loss (+ gradient) from precomputed margins — saves the A·w pass."""
+        return shard.eval_from_margins(margins, self.LOSS_TYPE, mask, need_grad)
+
     # --- MLlib per-example API parity (reference Gradient.compute) ---
     def compute(
         self,

@@ -28,6 +28,16 @@ class Updater:
     value of the regularization term at the new weights."""
 
     PROX_KIND: int = -1
+    #: True when the prox step is AFFINE in (w, g) — then margins propagate
+    #: algebraically (A·prox(w - s·g) is computable from A·w and A·g) and the
+    #: optimizer can run in margin-state-tracking mode (optimizer.py).
+    AFFINE_PROX: bool = False
+
+    def prox_margins(self, wm: torch.Tensor, gm: torch.Tensor, step: float,
+                     reg_param: float) -> torch.Tensor:
+        """Margins of the prox output from margins of (w, g); only valid when
+        AFFINE_PROX."""
+        raise NotImplementedError(f"{type(self).__name__} has no affine prox")
 
     def compute(
         self,
@@ -56,6 +66,10 @@ class SimpleUpdater(Updater):
     """w' = w - step*g; no regularization."""
 
     PROX_KIND = ops.PROX_SIMPLE
+    AFFINE_PROX = True
+
+    def prox_margins(self, wm, gm, step, reg_param):
+        return ops.axpby(1.0, wm, -step, gm)
 
 
 class L1Updater(Updater):
@@ -69,3 +83,7 @@ class SquaredL2Updater(Updater):
     """w' = w*(1 - step*lambda) - step*g; reg = (lambda/2) ||w'||^2."""
 
     PROX_KIND = ops.PROX_SQUARED_L2
+    AFFINE_PROX = True
+
+    def prox_margins(self, wm, gm, step, reg_param):
+        return ops.axpby(1.0 - step * reg_param, wm, -step, gm)

@@ -102,6 +102,33 @@ def csr_eval(
     return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, need_grad)
 
 
+def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    if _use_hip(features):
+        return _get_hip().dense_margins(features, v)
+    return reference.dense_margins(features, v)
+
+
+def dense_eval_from_margins(features, margins, labels, loss_type, mask=None, need_grad=True):
+    if _use_hip(features):
+        return _get_hip().dense_eval_from_margins(features, margins, labels, loss_type, mask, need_grad)
+    return reference.dense_eval_from_margins(features, margins, labels, loss_type, mask, need_grad)
+
+
+def csr_margins(rowptr, col, val, v):
+    if _use_hip(val):
+        return _get_hip().csr_margins(rowptr, col, val, v)
+    return reference.csr_margins(rowptr, col, val, v)
+
+
+def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
+                          mask=None, d=None, csc=None, need_grad=True):
+    if _use_hip(val):
+        return _get_hip().csr_eval_from_margins(rowptr, col, val, margins, labels,
+                                                loss_type, mask, d, csc, need_grad)
+    return reference.csr_eval_from_margins(rowptr, col, val, margins, labels,
+                                           loss_type, mask, d, csc, need_grad)
+
+
 def prox(
     kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float
 ) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -143,6 +170,10 @@ __all__ = [
     "PROX_SQUARED_L2",
     "dense_eval",
     "csr_eval",
+    "dense_margins",
+    "dense_eval_from_margins",
+    "csr_margins",
+    "csr_eval_from_margins",
     "prox",
     "axpby",
     "fused_scalars",

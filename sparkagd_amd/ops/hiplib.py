@@ -55,9 +55,9 @@ def load() -> ctypes.CDLL:
 
     P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
     lib.agd_dense_eval.restype = I
-    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, P]
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, I, P]
     lib.agd_csr_eval.restype = I
-    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, P]
+    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, I, P]
     lib.agd_axpby.restype = I
     lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
     lib.agd_prox.restype = I
@@ -135,7 +135,69 @@ def dense_eval(
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
         _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
         margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),  # nt A-stream: +6-9% measured
-        _stream(features),
+        0, _stream(features),
+    )
+    _check(rc)
+    return grad, loss_count
+
+
+def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """margins = A @ v for an arbitrary d-vector (margin-state tracking)."""
+    lib = load()
+    assert features.is_cuda and features.is_contiguous() and features.ndim == 2
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    acc = _ACC_DTYPE[features.dtype]
+    if v.dtype != acc:
+        raise TypeError(f"vector dtype {v.dtype} must be {acc}")
+    margins_algo = int(os.environ.get("SPARKAGD_MARGINS_ALGO", "0"))
+    n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype, margins_algo))
+    margins = torch.empty(n_slabs * n, dtype=acc, device=features.device)
+    rc = lib.agd_dense_eval(
+        _ptr(features), a_dtype, None, None, _ptr(v.contiguous()),
+        n, d, None, None, _ptr(margins), None, None, 1, 0, n_slabs, 0,
+        margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
+        1, _stream(features),
+    )
+    _check(rc)
+    return margins.narrow(0, 0, n)
+
+
+def dense_eval_from_margins(
+    features: torch.Tensor,
+    margins: torch.Tensor,
+    labels: torch.Tensor,
+    loss_type: int,
+    mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
+) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
+    """multiplier/loss (+ A^T·m when need_grad) from precomputed margins."""
+    lib = load()
+    assert features.is_cuda and features.is_contiguous() and features.ndim == 2
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    acc = _ACC_DTYPE[features.dtype]
+    assert margins.dtype == acc and margins.numel() == n
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, features.device)
+    dev = features.device
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    mult = torch.empty(n, dtype=acc, device=dev)
+    if need_grad:
+        grad = torch.empty(d, dtype=acc, device=dev)
+        n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
+        part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
+    else:
+        grad = part = None
+        n_rb = 1
+    rc = lib.agd_dense_eval(
+        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), None,
+        n, d, _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
+        _ptr(part), n_rb, loss_type, 1, 1 if need_grad else 0,
+        1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
+        2, _stream(features),
     )
     _check(rc)
     return grad, loss_count
@@ -188,7 +250,49 @@ def csr_eval(
         _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask),
         _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
         _ptr(margins), _ptr(mult), loss_type,
-        _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, _stream(val),
+        _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 0, _stream(val),
+    )
+    _check(rc)
+    return grad, loss_count
+
+
+def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
+    lib = load()
+    n = rowptr.numel() - 1
+    margins = torch.empty(n, dtype=torch.float32, device=val.device)
+    rc = lib.agd_csr_eval(
+        _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
+        None, None, _ptr(v.contiguous()), n, val.numel(), v.numel(),
+        None, None, _ptr(margins), None, 0, None, None, None, 0, 1, _stream(val),
+    )
+    _check(rc)
+    return margins
+
+
+def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
+                          mask=None, d=None, csc=None, need_grad=True):
+    lib = load()
+    n = rowptr.numel() - 1
+    d = d if d is not None else 0
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, val.device)
+    dev = val.device
+    if not need_grad:
+        grad = None; cp = cr = cv = None
+    elif csc is None:
+        grad = torch.zeros(d, dtype=torch.float32, device=dev); cp = cr = cv = None
+    else:
+        grad = torch.empty(d, dtype=torch.float32, device=dev); cp, cr, cv = csc
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    mult = torch.empty(n, dtype=torch.float32, device=dev)
+    rc = lib.agd_csr_eval(
+        _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
+        _ptr(labels), _ptr(mask), None, n, val.numel(), d,
+        _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
+        loss_type, _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 2,
+        _stream(val),
     )
     _check(rc)
     return grad, loss_count

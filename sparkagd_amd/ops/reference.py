@@ -128,6 +128,63 @@ def csr_eval(
     return grad_sum, loss_count
 
 
+def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """margins = A @ v (margin-state tracking support)."""
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    return features.to(acc) @ v.to(acc)
+
+
+def dense_eval_from_margins(
+    features: torch.Tensor,
+    margins: torch.Tensor,
+    labels: torch.Tensor,
+    loss_type: int,
+    mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
+) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    mult, loss = _multiplier_and_loss(margins, labels, loss_type)
+    if mask is not None:
+        m = mask.to(mult.dtype)
+        mult = mult * m
+        loss = loss * m
+        count = mask.sum().to(torch.float64)
+    else:
+        count = torch.tensor(float(features.shape[0]), dtype=torch.float64, device=features.device)
+    loss_count = torch.stack([loss.to(torch.float64).sum(), count])
+    if not need_grad:
+        return None, loss_count
+    grad_sum = (features.to(acc).T @ mult.to(acc))
+    return grad_sum, loss_count
+
+
+def csr_margins(rowptr, col, val, v: torch.Tensor, n: Optional[int] = None) -> torch.Tensor:
+    n = rowptr.numel() - 1
+    a = torch.sparse_csr_tensor(rowptr.to(torch.int64), col.to(torch.int64),
+                                val.to(torch.float32), size=(n, v.numel()))
+    return a @ v.to(torch.float32)
+
+
+def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
+                          mask=None, d=None, csc=None, need_grad=True):
+    n = rowptr.numel() - 1
+    mult, loss = _multiplier_and_loss(margins, labels, loss_type)
+    if mask is not None:
+        m = mask.to(mult.dtype)
+        mult = mult * m
+        loss = loss * m
+        count = mask.sum().to(torch.float64)
+    else:
+        count = torch.tensor(float(n), dtype=torch.float64, device=val.device)
+    loss_count = torch.stack([loss.to(torch.float64).sum(), count])
+    if not need_grad:
+        return None, loss_count
+    a = torch.sparse_csr_tensor(rowptr.to(torch.int64), col.to(torch.int64),
+                                val.to(torch.float32), size=(n, d))
+    grad_sum = a.t() @ mult.to(torch.float32)
+    return grad_sum, loss_count
+
+
 def prox(
     kind: int,
     w: torch.Tensor,

@@ -91,6 +91,8 @@ def run(
     resume_from: Optional[str] = None,
     iteration_hook=None,
     check_replication_every: int = 0,
+    track_margins: str | bool = "auto",
+    margin_refresh_every: int = 0,
 ) -> Tuple[torch.Tensor, List[float]]:
     """Run accelerated proximal gradient descent.
 
@@ -101,6 +103,25 @@ def run(
     ``iteration_hook(n_iter)`` runs at the end of each completed iteration;
     returning the string "stop" ends the loop (used by bench.py to bracket
     exactly K timed steps).
+
+    **Margin-state tracking** (``track_margins``, default auto): the same
+    linear-operator caching TFOCS itself performs. Margins are linear in the
+    weights, and for AFFINE prox operators (Simple/SquaredL2) the margins of
+    the AT iterates propagate algebraically:
+
+        A·y  = (1-θ)·A·x_old + θ·A·z_old            (free)
+        A·z' = prox-margins(A·z_old, A·g_y, step)   (free given A·g_y)
+        A·x' = (1-θ)·A·x_old + θ·A·z'               (free)
+
+    so the only data passes per accepted backtracking trial are the A^T·m
+    gradient pass and one margins pass over g_y — 2 instead of 3 (and a
+    rejected trial costs 2 instead of 3; loss-only f_x checks cost ZERO
+    passes). The math is identical up to fp accumulation order;
+    ``margin_refresh_every`` > 0 recomputes the tracked margins from the
+    weight vectors every k iterations to bound drift. Disabled automatically
+    for non-affine updaters (L1) and when ``resume_from``/bitwise
+    reproducibility against the non-tracking path is required
+    (``track_margins=False``).
     """
     comm = comm or Communicator()
     backtrack_tol = 1e-10
@@ -134,12 +155,39 @@ def run(
         eval_state["seconds"] += time.perf_counter() - t0
         return out
 
+    def apply_smooth_margins(vm, need_grad=True):
+        """applySmooth from tracked margins: multiplier/loss (+ A^T·m)."""
+        t0 = time.perf_counter()
+        grad_sum, loss_count = gradient.eval_from_margins(data, vm, need_grad=need_grad)
+        if grad_sum is not None:
+            comm.allreduce_(grad_sum)
+        comm.allreduce_(loss_count)
+        lc = loss_count.to("cpu")
+        loss_sum, count = float(lc[0]), float(lc[1])
+        if count > 0 and grad_sum is not None:
+            grad_sum.div_(count)
+        eval_state["n"] += 1
+        eval_state["seconds"] += time.perf_counter() - t0
+        return (loss_sum / count if count > 0 else float("nan")), grad_sum, count
+
+    # Margin-state tracking eligibility (see docstring).
+    tracking = bool(track_margins) and getattr(updater, "AFFINE_PROX", False) \
+        and hasattr(data, "margins") and hasattr(data, "eval_from_margins")
+    xm = zm = None
+    if tracking:
+        xm = gradient.margins(data, x)
+        zm = xm.clone() if torch.equal(x, z) else gradient.margins(data, z)
+
     broke = False
     for n_iter in range(start_iter, num_iterations + 1):
         t_iter0 = time.perf_counter()
         eval_n0, eval_s0 = eval_state["n"], eval_state["seconds"]
+        if tracking and margin_refresh_every > 0 and n_iter % margin_refresh_every == 0:
+            xm = gradient.margins(data, x)
+            zm = gradient.margins(data, z)
         # Auslender and Teboulle's accelerated method (AGD.scala:237-255).
         x_old, z_old = x, z
+        xm_old, zm_old = xm, zm
         L_old = L
         L = L * alpha
         theta_old = theta
@@ -155,10 +203,18 @@ def run(
             # after a restart gives theta = 1 (AGD.scala:248).
             theta = 2.0 / (1.0 + math.sqrt(1.0 + 4.0 * (L / L_old) / (theta_old * theta_old)))
             y = ops.axpby(1.0 - theta, x_old, theta, z_old)
-            f_y, g_y, _count = apply_smooth(y)
+            if tracking:
+                ym = ops.axpby(1.0 - theta, xm_old, theta, zm_old)
+                f_y, g_y, _count = apply_smooth_margins(ym)  # A^T·m pass only
+            else:
+                f_y, g_y, _count = apply_smooth(y)
             step = 1.0 / (theta * L)
             z, _ = updater.compute(z_old, g_y, step, 1, reg_param)
             x = ops.axpby(1.0 - theta, x_old, theta, z)
+            if tracking:
+                gm = gradient.margins(data, g_y)  # the only other data pass
+                zm = updater.prox_margins(zm_old, gm, step, reg_param)
+                xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
 
             if beta >= 1.0:
                 scal = None  # computed after the loop for the convergence test
@@ -172,14 +228,21 @@ def run(
                 break
 
             if backtrack_simple:
-                # the simple test needs f_x only: loss-only pass (no A^T·m)
-                f_x, _gx_none, _ = apply_smooth(x, need_grad=False)
+                # the simple test needs f_x only: loss-only evaluation
+                # (tracking: ZERO data passes — f_x comes from xm)
+                if tracking:
+                    f_x, _gx_none, _ = apply_smooth_margins(xm, need_grad=False)
+                else:
+                    f_x, _gx_none, _ = apply_smooth(x, need_grad=False)
                 f_x_bt = f_x
                 q_x = f_y + float(scal[1]) + 0.5 * L * xy_sq
                 localL = L + 2.0 * max(f_x - q_x, 0.0) / xy_sq
                 backtrack_simple = abs(f_y - f_x) >= backtrack_tol * max(abs(f_x), abs(f_y))
             else:
-                f_x, g_x, _ = apply_smooth(x)
+                if tracking:
+                    f_x, g_x, _ = apply_smooth_margins(xm)
+                else:
+                    f_x, g_x, _ = apply_smooth(x)
                 f_x_bt = f_x
                 localL = 2.0 * float(ops.dot_diff(x, y, g_x, g_y)) / xy_sq
 
@@ -200,7 +263,10 @@ def run(
         # extra full-data pass at x (TFOCS validation); 'backtrack' reuses the
         # accepted backtracking evaluation; 'none' records f_y + c_y.
         if loss_history_mode == "exact":
-            f_x2, _g_x2, _ = apply_smooth(x)
+            if tracking:
+                f_x2, _g_x2, _ = apply_smooth_margins(xm, need_grad=False)
+            else:
+                f_x2, _g_x2, _ = apply_smooth(x, need_grad=False)
             c_x = float(updater.reg_value(x, reg_param))
             loss_history.append(f_x2 + c_x)
         elif loss_history_mode == "backtrack" and f_x_bt is not None:
@@ -227,6 +293,8 @@ def run(
         # Gradient-test restart (O'Donoghue & Candès 2013; AGD.scala:326-331).
         if not broke and may_restart and float(scal[4]) > 0.0:
             z = x.clone()
+            if tracking:
+                zm = xm.clone()
             theta = math.inf
             backtrack_simple = True
             restarted = True
