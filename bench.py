@@ -10,12 +10,13 @@ AGD iterations, then EXACTLY K timed iterations bracketed by a barrier +
 torch.cuda.synchronize() on both sides; elapsed is the MAX over ranks; rank 0
 prints ONE JSON line.
 
-Metric: examples/sec — examples processed through the gradient pipeline
-(evaluations x rows) per second, aggregated over all ranks. Each AGD iteration
-performs 2 full-data evaluations in the default backtracking configuration
-(eval at y + the accepted backtracking eval at x; the reference's third
+Metric: examples/sec — loss/gradient evaluations x rows per second,
+aggregated over all ranks. Each AGD iteration performs 2 evaluations (at y,
+and the accepted backtracking trial at x; the reference's third
 TFOCS-validation pass, AGD.scala:302-307, is off — loss history reuses the
-accepted f_x). Weak scaling: rows-per-GPU fixed as N grows.
+accepted f_x) over 2 physical shard streams (margin-state tracking;
+config.data_passes_per_step reports the honest pass count). Weak scaling:
+rows-per-GPU fixed as N grows.
 """
 
 from __future__ import annotations
