@@ -78,10 +78,14 @@ __device__ __forceinline__ T wave_reduce_sum(T v) {
   return v;
 }
 
-// Block-level fp64 reduction of NACC accumulators, then one atomicAdd each.
+// Block-level fp64 reduction of NACC accumulators into a per-block partial
+// slot; a one-block k_reduce_partials finishes the sum. Two stages instead
+// of leader atomics: 8192 same-word fp64 atomics serialize to ~0.5 ms
+// (measured, d=1e7 fused-scalars) and are order-nondeterministic; the
+// partial buffer costs ~30 us and is bitwise reproducible.
 template <int NACC>
-__device__ __forceinline__ void block_reduce_atomic(double (&acc)[NACC],
-                                                    double* out) {
+__device__ __forceinline__ void block_reduce_partial(double (&acc)[NACC],
+                                                     double* part) {
   __shared__ double lds[WAVES_PER_BLOCK][NACC];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
@@ -98,7 +102,39 @@ __device__ __forceinline__ void block_reduce_atomic(double (&acc)[NACC],
 #pragma unroll
       for (int off = WAVES_PER_BLOCK / 2; off > 0; off >>= 1)
         v += __shfl_xor(v, off, WAVE);
-      if (lane == 0 && v != 0.0) atomicAdd(&out[k], v);
+      if (lane == 0) part[(ll)blockIdx.x * NACC + k] = v;
+    }
+  }
+}
+
+// Final stage: one block sums the per-block partials into out[NACC]
+// (ACCUMULATES into out so callers may chain; callers zero out first).
+template <int NACC>
+__global__ __launch_bounds__(BLOCK) void k_reduce_partials(
+    const double* __restrict__ part, int nblocks, double* __restrict__ out) {
+  double acc[NACC];
+#pragma unroll
+  for (int k = 0; k < NACC; ++k) acc[k] = 0.0;
+  for (int i = threadIdx.x; i < nblocks; i += BLOCK)
+#pragma unroll
+    for (int k = 0; k < NACC; ++k) acc[k] += part[(ll)i * NACC + k];
+  __shared__ double lds[WAVES_PER_BLOCK][NACC];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+#pragma unroll
+  for (int k = 0; k < NACC; ++k) {
+    acc[k] = wave_reduce_sum(acc[k]);
+    if (lane == 0) lds[wid][k] = acc[k];
+  }
+  __syncthreads();
+  if (wid == 0 && lane < WAVES_PER_BLOCK) {
+#pragma unroll
+    for (int k = 0; k < NACC; ++k) {
+      double v = lds[lane][k];
+#pragma unroll
+      for (int off = WAVES_PER_BLOCK / 2; off > 0; off >>= 1)
+        v += __shfl_xor(v, off, WAVE);
+      if (lane == 0) out[k] += v;
     }
   }
 }
@@ -377,7 +413,7 @@ template <typename TACC>
 __global__ __launch_bounds__(BLOCK) void k_multiplier(
     const TACC* __restrict__ margins, const float* __restrict__ labels,
     const unsigned char* __restrict__ mask, int loss_type, ll n, int n_slabs,
-    TACC* __restrict__ mult, double* __restrict__ loss_count) {
+    TACC* __restrict__ mult, double* __restrict__ red_part) {
   double lsum = 0.0, cnt = 0.0;
   const ll stride = (ll)gridDim.x * BLOCK;
   for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
@@ -407,7 +443,7 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
     cnt += 1.0;
   }
   double acc[2] = {lsum, cnt};
-  block_reduce_atomic<2>(acc, loss_count);
+  block_reduce_partial<2>(acc, red_part);
 }
 
 // ---------------------------------------------------------------------------
@@ -595,7 +631,8 @@ __global__ __launch_bounds__(BLOCK) void k_prox(int kind, const T* __restrict__ 
                                                 const T* __restrict__ g,
                                                 double step, double lam,
                                                 T* __restrict__ out,
-                                                double* __restrict__ reg, ll n) {
+                                                double* __restrict__ reg_part,
+                                                ll n) {
   constexpr int VE = 16 / sizeof(T);
   const ll stride = (ll)gridDim.x * BLOCK;
   const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
@@ -614,7 +651,7 @@ __global__ __launch_bounds__(BLOCK) void k_prox(int kind, const T* __restrict__ 
   if (t < n) out[t] = prox_elem(kind, w[t], g[t], ts, tl, racc);
   if (kind != PROX_SIMPLE) {
     double acc[1] = {racc};
-    block_reduce_atomic<1>(acc, reg);
+    block_reduce_partial<1>(acc, reg_part);
   }
 }
 
@@ -641,7 +678,7 @@ __device__ __forceinline__ void fused_scalars_elem(double xi, double yi,
 template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_fused_scalars(
     const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ gy,
-    const T* __restrict__ xold, double* __restrict__ out, ll n) {
+    const T* __restrict__ xold, double* __restrict__ out_part, ll n) {
   constexpr int VE = 16 / sizeof(T);
   double acc[5] = {0, 0, 0, 0, 0};
   const ll stride = (ll)gridDim.x * BLOCK;
@@ -659,13 +696,13 @@ __global__ __launch_bounds__(BLOCK) void k_fused_scalars(
   }
   const ll t = nv * VE + gid;
   if (t < n) fused_scalars_elem<T>(x[t], y[t], gy[t], xold[t], acc);
-  block_reduce_atomic<5>(acc, out);
+  block_reduce_partial<5>(acc, out_part);
 }
 
 template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_dot_diff(
     const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ gx,
-    const T* __restrict__ gy, double* __restrict__ out, ll n) {
+    const T* __restrict__ gy, double* __restrict__ out_part, ll n) {
   constexpr int VE = 16 / sizeof(T);
   double a = 0;
   const ll stride = (ll)gridDim.x * BLOCK;
@@ -685,7 +722,7 @@ __global__ __launch_bounds__(BLOCK) void k_dot_diff(
   if (t < n)
     a += ((double)x[t] - (double)y[t]) * ((double)gx[t] - (double)gy[t]);
   double acc[1] = {a};
-  block_reduce_atomic<1>(acc, out);
+  block_reduce_partial<1>(acc, out_part);
 }
 
 // ---------------------------------------------------------------------------
@@ -770,7 +807,8 @@ static int dense_eval_t(const void* A, const float* labels,
                         void* grad_out, double* loss_count, void* margins_ws,
                         void* mult_ws, void* part_ws, ll n_rb, int loss_type,
                         int n_slabs, int need_grad, int margins_algo,
-                        int nt_loads, int mode, hipStream_t stream) {
+                        int nt_loads, int mode, double* red_ws,
+                        hipStream_t stream) {
   const TA* a = (const TA*)A;
   const TACC* wp = (const TACC*)w;
   TACC* margins = (TACC*)margins_ws;
@@ -822,7 +860,9 @@ static int dense_eval_t(const void* A, const float* labels,
     const int grid = grid_for(n, BLOCK);
     hipLaunchKernelGGL((k_multiplier<TACC>), dim3(grid), dim3(BLOCK), 0, stream,
                        margins, labels, mask, loss_type, n, n_slabs, mult,
-                       loss_count);
+                       red_ws);
+    hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, stream,
+                       red_ws, grid, loss_count);
   }
   if (need_grad) {
     {
@@ -855,19 +895,19 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
                               void* margins_ws, void* mult_ws, void* part_ws,
                               long long n_rb, int loss_type, int n_slabs,
                               int need_grad, int margins_algo, int nt_loads,
-                              int mode, void* stream) {
+                              int mode, void* red_ws, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const float* lab = (const float*)labels;
   const unsigned char* msk = (const unsigned char*)mask;
   double* lc = (double*)loss_count;
   const int W = pick_w(a_dtype, d);
   switch (a_dtype * 10 + (W > 1 ? 1 : 0)) {
-    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
-    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
-    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
-    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
-    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
-    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, s);
+    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;
@@ -886,7 +926,7 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
                             void* mult_ws, int loss_type,
                             const void* csc_colptr, const void* csc_row,
                             const void* csc_val, int need_grad, int mode,
-                            void* stream) {
+                            void* red_ws, void* stream) {
   (void)nnz;
   hipStream_t s = (hipStream_t)stream;
   const int* rp = (const int*)rowptr;
@@ -908,7 +948,9 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
     hipLaunchKernelGGL((k_multiplier<float>), dim3(grid), dim3(BLOCK), 0, s,
                        margins, (const float*)labels,
                        (const unsigned char*)mask, loss_type, n, 1, mult,
-                       (double*)loss_count);
+                       (double*)red_ws);
+    hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, s,
+                       (double*)red_ws, grid, (double*)loss_count);
   }
   if (!need_grad) {
     // loss-only evaluation (simple-backtracking f_x trials): skip A^T.m
@@ -947,21 +989,24 @@ extern "C" int agd_axpby(double a, const void* x, double b, const void* y,
 // reg (double[1]) must be zeroed by the caller.
 extern "C" int agd_prox(int kind, const void* w, const void* g, double step,
                         double lam, void* out, void* reg, long long n,
-                        int dtype, void* stream) {
+                        int dtype, void* red_ws, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const int grid = grid_for(n, BLOCK * 4);
   if (dtype == 1)
     hipLaunchKernelGGL((k_prox<float>), dim3(grid), dim3(BLOCK), 0, s, kind,
                        (const float*)w, (const float*)g, step, lam, (float*)out,
-                       (double*)reg, n);
+                       (double*)red_ws, n);
   else if (dtype == 2)
     hipLaunchKernelGGL((k_prox<double>), dim3(grid), dim3(BLOCK), 0, s, kind,
                        (const double*)w, (const double*)g, step, lam,
-                       (double*)out, (double*)reg, n);
+                       (double*)out, (double*)red_ws, n);
   else {
     snprintf(g_err, sizeof(g_err), "agd_prox: bad dtype %d", dtype);
     return 2;
   }
+  if (kind != PROX_SIMPLE)
+    hipLaunchKernelGGL((k_reduce_partials<1>), dim3(1), dim3(BLOCK), 0, s,
+                       (double*)red_ws, grid, (double*)reg);
   HIP_CHECK(hipGetLastError());
   return 0;
 }
@@ -969,21 +1014,23 @@ extern "C" int agd_prox(int kind, const void* w, const void* g, double step,
 // out (double[5]) must be zeroed by the caller.
 extern "C" int agd_fused_scalars(const void* x, const void* y, const void* gy,
                                  const void* xold, void* out, long long n,
-                                 int dtype, void* stream) {
+                                 int dtype, void* red_ws, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const int grid = grid_for(n, BLOCK * 4);
   if (dtype == 1)
     hipLaunchKernelGGL((k_fused_scalars<float>), dim3(grid), dim3(BLOCK), 0, s,
                        (const float*)x, (const float*)y, (const float*)gy,
-                       (const float*)xold, (double*)out, n);
+                       (const float*)xold, (double*)red_ws, n);
   else if (dtype == 2)
     hipLaunchKernelGGL((k_fused_scalars<double>), dim3(grid), dim3(BLOCK), 0, s,
                        (const double*)x, (const double*)y, (const double*)gy,
-                       (const double*)xold, (double*)out, n);
+                       (const double*)xold, (double*)red_ws, n);
   else {
     snprintf(g_err, sizeof(g_err), "agd_fused_scalars: bad dtype %d", dtype);
     return 2;
   }
+  hipLaunchKernelGGL((k_reduce_partials<5>), dim3(1), dim3(BLOCK), 0, s,
+                     (double*)red_ws, grid, (double*)out);
   HIP_CHECK(hipGetLastError());
   return 0;
 }
@@ -991,21 +1038,23 @@ extern "C" int agd_fused_scalars(const void* x, const void* y, const void* gy,
 // out (double[1]) must be zeroed by the caller.
 extern "C" int agd_dot_diff(const void* x, const void* y, const void* gx,
                             const void* gy, void* out, long long n, int dtype,
-                            void* stream) {
+                            void* red_ws, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const int grid = grid_for(n, BLOCK * 4);
   if (dtype == 1)
     hipLaunchKernelGGL((k_dot_diff<float>), dim3(grid), dim3(BLOCK), 0, s,
                        (const float*)x, (const float*)y, (const float*)gx,
-                       (const float*)gy, (double*)out, n);
+                       (const float*)gy, (double*)red_ws, n);
   else if (dtype == 2)
     hipLaunchKernelGGL((k_dot_diff<double>), dim3(grid), dim3(BLOCK), 0, s,
                        (const double*)x, (const double*)y, (const double*)gx,
-                       (const double*)gy, (double*)out, n);
+                       (const double*)gy, (double*)red_ws, n);
   else {
     snprintf(g_err, sizeof(g_err), "agd_dot_diff: bad dtype %d", dtype);
     return 2;
   }
+  hipLaunchKernelGGL((k_reduce_partials<1>), dim3(1), dim3(BLOCK), 0, s,
+                     (double*)red_ws, grid, (double*)out);
   HIP_CHECK(hipGetLastError());
   return 0;
 }

@@ -55,17 +55,17 @@ def load() -> ctypes.CDLL:
 
     P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
     lib.agd_dense_eval.restype = I
-    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, I, P]
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, I, P, P]
     lib.agd_csr_eval.restype = I
-    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, I, P]
+    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, I, P, P]
     lib.agd_axpby.restype = I
     lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
     lib.agd_prox.restype = I
-    lib.agd_prox.argtypes = [I, P, P, D, D, P, P, LL, I, P]
+    lib.agd_prox.argtypes = [I, P, P, D, D, P, P, LL, I, P, P]
     lib.agd_fused_scalars.restype = I
-    lib.agd_fused_scalars.argtypes = [P, P, P, P, P, LL, I, P]
+    lib.agd_fused_scalars.argtypes = [P, P, P, P, P, LL, I, P, P]
     lib.agd_dot_diff.restype = I
-    lib.agd_dot_diff.argtypes = [P, P, P, P, P, LL, I, P]
+    lib.agd_dot_diff.argtypes = [P, P, P, P, P, LL, I, P, P]
 
     _lib = lib
     return lib
@@ -84,6 +84,19 @@ def _ptr(t: Optional[torch.Tensor]) -> Optional[ctypes.c_void_p]:
     if t is None:
         return None
     return ctypes.c_void_p(t.data_ptr())
+
+
+_red_ws_cache = {}
+
+
+def _red_ws(device) -> torch.Tensor:
+    """Per-device fp64 partials buffer for the two-stage scalar reductions
+    (max grid 8192 x up to 5 accumulators)."""
+    t = _red_ws_cache.get(device)
+    if t is None:
+        t = torch.empty(8192 * 5, dtype=torch.float64, device=device)
+        _red_ws_cache[device] = t
+    return t
 
 
 def _prep_mask(mask: Optional[torch.Tensor], device) -> Optional[torch.Tensor]:
@@ -135,7 +148,7 @@ def dense_eval(
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
         _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
         margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),  # nt A-stream: +6-9% measured
-        0, _stream(features),
+        0, _ptr(_red_ws(features.device)), _stream(features),
     )
     _check(rc)
     return grad, loss_count
@@ -157,7 +170,7 @@ def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
         _ptr(features), a_dtype, None, None, _ptr(v.contiguous()),
         n, d, None, None, _ptr(margins), None, None, 1, 0, n_slabs, 0,
         margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
-        1, _stream(features),
+        1, _ptr(_red_ws(features.device)), _stream(features),
     )
     _check(rc)
     return margins.narrow(0, 0, n)
@@ -197,7 +210,7 @@ def dense_eval_from_margins(
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
         _ptr(part), n_rb, loss_type, 1, 1 if need_grad else 0,
         1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
-        2, _stream(features),
+        2, _ptr(_red_ws(features.device)), _stream(features),
     )
     _check(rc)
     return grad, loss_count
@@ -250,7 +263,8 @@ def csr_eval(
         _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask),
         _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
         _ptr(margins), _ptr(mult), loss_type,
-        _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 0, _stream(val),
+        _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 0,
+        _ptr(_red_ws(val.device)), _stream(val),
     )
     _check(rc)
     return grad, loss_count
@@ -263,7 +277,8 @@ def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
     rc = lib.agd_csr_eval(
         _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
         None, None, _ptr(v.contiguous()), n, val.numel(), v.numel(),
-        None, None, _ptr(margins), None, 0, None, None, None, 0, 1, _stream(val),
+        None, None, _ptr(margins), None, 0, None, None, None, 0, 1,
+        _ptr(_red_ws(val.device)), _stream(val),
     )
     _check(rc)
     return margins
@@ -292,7 +307,7 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
         _ptr(labels), _ptr(mask), None, n, val.numel(), d,
         _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
         loss_type, _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 2,
-        _stream(val),
+        _ptr(_red_ws(val.device)), _stream(val),
     )
     _check(rc)
     return grad, loss_count
@@ -316,7 +331,8 @@ def prox(kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float) -
     out = torch.empty_like(w)
     reg = torch.zeros((), dtype=torch.float64, device=w.device)
     rc = lib.agd_prox(kind, _ptr(w.contiguous()), _ptr(g.contiguous()), float(step),
-                      float(lam), _ptr(out), _ptr(reg), w.numel(), _VEC_DTYPE[w.dtype], _stream(w))
+                      float(lam), _ptr(out), _ptr(reg), w.numel(), _VEC_DTYPE[w.dtype],
+                      _ptr(_red_ws(w.device)), _stream(w))
     _check(rc)
     return out, reg
 
@@ -326,7 +342,8 @@ def fused_scalars(x: torch.Tensor, y: torch.Tensor, g_y: torch.Tensor, x_old: to
     out = torch.zeros(5, dtype=torch.float64, device=x.device)
     rc = lib.agd_fused_scalars(_ptr(x.contiguous()), _ptr(y.contiguous()),
                                _ptr(g_y.contiguous()), _ptr(x_old.contiguous()),
-                               _ptr(out), x.numel(), _VEC_DTYPE[x.dtype], _stream(x))
+                               _ptr(out), x.numel(), _VEC_DTYPE[x.dtype],
+                               _ptr(_red_ws(x.device)), _stream(x))
     _check(rc)
     return out
 
@@ -336,6 +353,7 @@ def dot_diff(x: torch.Tensor, y: torch.Tensor, g_x: torch.Tensor, g_y: torch.Ten
     out = torch.zeros((), dtype=torch.float64, device=x.device)
     rc = lib.agd_dot_diff(_ptr(x.contiguous()), _ptr(y.contiguous()),
                           _ptr(g_x.contiguous()), _ptr(g_y.contiguous()),
-                          _ptr(out), x.numel(), _VEC_DTYPE[x.dtype], _stream(x))
+                          _ptr(out), x.numel(), _VEC_DTYPE[x.dtype],
+                          _ptr(_red_ws(x.device)), _stream(x))
     _check(rc)
     return out
