@@ -532,10 +532,31 @@ __global__ __launch_bounds__(BLOCK) void k_csr_margins(
   const int wid = threadIdx.x / WAVE;
   const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
   const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  using i32x4 = __attribute__((ext_vector_type(4))) int;
   for (ll r = wave_gid; r < n; r += n_waves) {
     const int k_lo = rowptr[r], k_hi = rowptr[r + 1];
     float acc = 0.f;
-    for (int k = k_lo + lane; k < k_hi; k += WAVE) acc += val[k] * w[col[k]];
+    // 16-B val/col loads (4 nnz per lane); scalar head to reach 16-B
+    // alignment, scalar tail for the remainder. The w gather stays 4 B
+    // (random within w — L2/L3-served).
+    const int head_end = (k_lo + 3) & ~3;
+    const int body_end = k_hi & ~3;
+    if (head_end + 4 * WAVE <= body_end) {
+      for (int k = k_lo + lane; k < head_end; k += WAVE) acc += val[k] * w[col[k]];
+      for (int k = head_end + 4 * lane; k + 4 <= body_end; k += 4 * WAVE) {
+        const f32x4 v = *(const f32x4*)&val[k];
+        const i32x4 c = *(const i32x4*)&col[k];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc += v[j] * w[c[j]];
+      }
+      // every full 4-group in [head_end, body_end) is covered by the strided
+      // group loop above; the tail is only [body_end, k_hi) (< 4 elements)
+      for (int k = body_end + lane; k < k_hi; k += WAVE)
+        acc += val[k] * w[col[k]];
+    } else {
+      for (int k = k_lo + lane; k < k_hi; k += WAVE) acc += val[k] * w[col[k]];
+    }
     acc = wave_reduce_sum(acc);
     if (lane == 0) margins[r] = acc;
   }
