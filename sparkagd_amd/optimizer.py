@@ -21,6 +21,7 @@ from __future__ import annotations
 
 import logging
 import math
+import os
 import time
 from typing import List, Optional, Tuple
 
@@ -33,6 +34,9 @@ from .models.updater import Updater
 from .parallel.comm import Communicator
 
 logger = logging.getLogger(__name__)
+
+#: cumulative all-reduce wall time when SPARKAGD_TIMING_DETAIL=1
+_COMM_SECONDS = [0.0]
 
 
 class Optimizer:
@@ -57,11 +61,23 @@ def _apply_smooth(
     ``need_grad=False`` evaluates the loss side only (one data pass instead
     of two): used for the simple-backtracking f_x trials, which the reference
     pays a full gradient evaluation for (AGD.scala:269) without ever using
-    g_x in the simple test."""
+    g_x in the simple test.
+
+    SPARKAGD_TIMING_DETAIL=1 splits compute vs all-reduce wall time into
+    _COMM_SECONDS (adds one device sync per eval — diagnostics only)."""
     grad_sum, loss_count = gradient.eval(data, v, mask, need_grad)
+    detail = os.environ.get("SPARKAGD_TIMING_DETAIL") == "1"
+    if detail:
+        if v.is_cuda:
+            torch.cuda.synchronize(v.device)
+        t_comm = time.perf_counter()
     if grad_sum is not None:
         comm.allreduce_(grad_sum)
     comm.allreduce_(loss_count)
+    if detail:
+        if v.is_cuda:
+            torch.cuda.synchronize(v.device)
+        _COMM_SECONDS[0] += time.perf_counter() - t_comm
     lc = loss_count.to("cpu", non_blocking=False)  # single host sync per eval
     loss_sum, count = float(lc[0]), float(lc[1])
     if count > 0 and grad_sum is not None:
