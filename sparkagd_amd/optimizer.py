@@ -175,9 +175,18 @@ def run(
         """applySmooth from tracked margins: multiplier/loss (+ A^T·m)."""
         t0 = time.perf_counter()
         grad_sum, loss_count = gradient.eval_from_margins(data, vm, need_grad=need_grad)
+        detail = os.environ.get("SPARKAGD_TIMING_DETAIL") == "1"
+        if detail:
+            if vm.is_cuda:
+                torch.cuda.synchronize(vm.device)
+            t_comm = time.perf_counter()
         if grad_sum is not None:
             comm.allreduce_(grad_sum)
         comm.allreduce_(loss_count)
+        if detail:
+            if vm.is_cuda:
+                torch.cuda.synchronize(vm.device)
+            _COMM_SECONDS[0] += time.perf_counter() - t_comm
         lc = loss_count.to("cpu")
         loss_sum, count = float(lc[0]), float(lc[1])
         if count > 0 and grad_sum is not None:
@@ -198,6 +207,7 @@ def run(
     for n_iter in range(start_iter, num_iterations + 1):
         t_iter0 = time.perf_counter()
         eval_n0, eval_s0 = eval_state["n"], eval_state["seconds"]
+        comm_s0 = _COMM_SECONDS[0]
         if tracking and margin_refresh_every > 0 and n_iter % margin_refresh_every == 0:
             xm = gradient.margins(data, x)
             zm = gradient.margins(data, z)
@@ -338,6 +348,10 @@ def run(
                 iter_seconds=time.perf_counter() - t_iter0,
                 n_evals=eval_state["n"] - eval_n0,
                 eval_seconds=eval_state["seconds"] - eval_s0,
+                **(
+                    {"comm_seconds": _COMM_SECONDS[0] - comm_s0}
+                    if os.environ.get("SPARKAGD_TIMING_DETAIL") == "1" else {}
+                ),
             )
 
         if (
