@@ -94,6 +94,8 @@ def main() -> int:
     p.add_argument("--csr", action="store_true", help="CSR-sparse shard instead of dense")
     p.add_argument("--nnz-per-row", type=int, default=64)
     p.add_argument("--eps", type=float, default=1e-3, help="relative loss-improvement epsilon for iters-to-eps")
+    p.add_argument("--solver", type=str, default="direct", choices=["direct", "gram"],
+                   help="gram = dual-space solver (K=A.A^T precompute; O(n_local*n_global) iterations)")
     args = p.parse_args()
 
     comm = init_from_env()
@@ -162,6 +164,7 @@ def main() -> int:
         loss_history_mode="backtrack",
         comm=comm,
         iteration_hook=hook,
+        solver=args.solver,
     )
 
     elapsed_local = state["t1"] - state["t0"]
@@ -208,6 +211,7 @@ def main() -> int:
                 "rows_per_gpu": args.rows,
                 "global_rows": global_rows,
                 "parallelism": f"dp{world}",
+                "solver": args.solver,
                 "evals_per_step": evals / max(timed_iters, 1),
                 "data_passes_per_step": passes / max(timed_iters, 1),
                 "weights_dtype": str(wdtype).replace("torch.", ""),

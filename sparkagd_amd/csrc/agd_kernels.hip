@@ -840,7 +840,7 @@ static int dense_eval_t(const void* A, const float* labels,
   const bool use_mfma = (algo == 2) && (sizeof(TA) == 2) && (W == 8);
 
   ll slab_w = d;
-  if (mode == 2) n_slabs = 1;  // margins provided by the caller
+  if (mode == 2 || mode == 3) n_slabs = 1;  // margins/mult provided by caller
   if (n_slabs > 1) {
     slab_w = (d + n_slabs - 1) / n_slabs;
     const ll align = (ll)WAVE * W;
@@ -848,7 +848,7 @@ static int dense_eval_t(const void* A, const float* labels,
     n_slabs = (int)((d + slab_w - 1) / slab_w);
   }
 
-  if (mode == 2) {
+  if (mode == 2 || mode == 3) {
     // skip the margins pass entirely
   } else if (use_mfma) {
     const ll tasks = ((n + MFMA_ROWS - 1) / MFMA_ROWS) * n_slabs;
@@ -877,7 +877,7 @@ static int dense_eval_t(const void* A, const float* labels,
     HIP_CHECK(hipGetLastError());
     return 0;
   }
-  {
+  if (mode != 3) {  // mode 3: the caller provides the multiplier vector
     const int grid = grid_for(n, BLOCK);
     hipLaunchKernelGGL((k_multiplier<TACC>), dim3(grid), dim3(BLOCK), 0, stream,
                        margins, labels, mask, loss_type, n, n_slabs, mult,

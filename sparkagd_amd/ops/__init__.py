@@ -114,6 +114,18 @@ def dense_eval_from_margins(features, margins, labels, loss_type, mask=None, nee
     return reference.dense_eval_from_margins(features, margins, labels, loss_type, mask, need_grad)
 
 
+def dense_multiplier_loss(features, margins, labels, loss_type, mask=None):
+    if _use_hip(features):
+        return _get_hip().dense_multiplier_loss(features, margins, labels, loss_type, mask)
+    return reference.dense_multiplier_loss(features, margins, labels, loss_type, mask)
+
+
+def dense_grad_from_mult(features, mult):
+    if _use_hip(features):
+        return _get_hip().dense_grad_from_mult(features, mult)
+    return reference.dense_grad_from_mult(features, mult)
+
+
 def csr_margins(rowptr, col, val, v):
     if _use_hip(val):
         return _get_hip().csr_margins(rowptr, col, val, v)
@@ -171,6 +183,8 @@ __all__ = [
     "dense_eval",
     "csr_eval",
     "dense_margins",
+    "dense_multiplier_loss",
+    "dense_grad_from_mult",
     "dense_eval_from_margins",
     "csr_margins",
     "csr_eval_from_margins",

@@ -216,6 +216,61 @@ def dense_eval_from_margins(
     return grad, loss_count
 
 
+def dense_multiplier_loss(
+    features: torch.Tensor,
+    margins: torch.Tensor,
+    labels: torch.Tensor,
+    loss_type: int,
+    mask: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(mult, loss_count) from precomputed margins — the Gram solver's
+    n-space evaluation (no data pass at all)."""
+    lib = load()
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    acc = _ACC_DTYPE[features.dtype]
+    assert margins.dtype == acc and margins.numel() == n
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, features.device)
+    dev = features.device
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    mult = torch.empty(n, dtype=acc, device=dev)
+    rc = lib.agd_dense_eval(
+        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), None,
+        n, d, None, _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
+        None, 1, loss_type, 1, 0,
+        1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
+        2, _ptr(_red_ws(dev)), _stream(features),
+    )
+    _check(rc)
+    return mult, loss_count
+
+
+def dense_grad_from_mult(features: torch.Tensor, mult: torch.Tensor) -> torch.Tensor:
+    """A^T @ mult for an arbitrary multiplier vector (Gram solver's final
+    weight materialization)."""
+    lib = load()
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    acc = _ACC_DTYPE[features.dtype]
+    assert mult.dtype == acc and mult.numel() == n
+    dev = features.device
+    grad = torch.empty(d, dtype=acc, device=dev)
+    n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
+    part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
+    rc = lib.agd_dense_eval(
+        _ptr(features), a_dtype, None, None, None,
+        n, d, _ptr(grad), None, None, _ptr(mult.contiguous()),
+        _ptr(part), n_rb, 0, 1, 1,
+        1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
+        3, _ptr(_red_ws(dev)), _stream(features),
+    )
+    _check(rc)
+    return grad
+
+
 def csr_eval(
     rowptr: torch.Tensor,
     col: torch.Tensor,

@@ -158,6 +158,29 @@ def dense_eval_from_margins(
     return grad_sum, loss_count
 
 
+def dense_multiplier_loss(
+    features: torch.Tensor,
+    margins: torch.Tensor,
+    labels: torch.Tensor,
+    loss_type: int,
+    mask: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    mult, loss = _multiplier_and_loss(margins, labels, loss_type)
+    if mask is not None:
+        m = mask.to(mult.dtype)
+        mult = mult * m
+        loss = loss * m
+        count = mask.sum().to(torch.float64)
+    else:
+        count = torch.tensor(float(features.shape[0]), dtype=torch.float64, device=features.device)
+    return mult, torch.stack([loss.to(torch.float64).sum(), count])
+
+
+def dense_grad_from_mult(features: torch.Tensor, mult: torch.Tensor) -> torch.Tensor:
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    return features.to(acc).T @ mult.to(acc)
+
+
 def csr_margins(rowptr, col, val, v: torch.Tensor, n: Optional[int] = None) -> torch.Tensor:
     n = rowptr.numel() - 1
     a = torch.sparse_csr_tensor(rowptr.to(torch.int64), col.to(torch.int64),

@@ -109,6 +109,8 @@ def run(
     check_replication_every: int = 0,
     track_margins: str | bool = "auto",
     margin_refresh_every: int = 0,
+    solver: str = "direct",
+    gram_op=None,
 ) -> Tuple[torch.Tensor, List[float]]:
     """Run accelerated proximal gradient descent.
 
@@ -139,6 +141,23 @@ def run(
     reproducibility against the non-tracking path is required
     (``track_margins=False``).
     """
+    if solver not in ("direct", "gram"):
+        raise ValueError("solver must be 'direct' or 'gram'")
+    if solver == "gram":
+        # Dual-space solver (gram.py): O(n_local * n_global) iterations for
+        # the n << d regime. Checkpoint/resume and mini-batching stay on the
+        # direct path.
+        from .gram import run_gram
+
+        if resume_from is not None or checkpoint_path is not None:
+            raise ValueError("checkpoint/resume requires the direct solver")
+        return run_gram(
+            data, gradient, updater, convergence_tol, num_iterations,
+            reg_param, initial_weights, L0, Lexact, beta, alpha, may_restart,
+            loss_history_mode=loss_history_mode, comm=comm, metrics=metrics,
+            iteration_hook=iteration_hook, gram_op=gram_op,
+        )
+
     comm = comm or Communicator()
     backtrack_tol = 1e-10
 

@@ -1,0 +1,87 @@
+"""Dual-space (Gram) solver: trajectory identity with the direct solver."""
+
+import math
+
+import pytest
+import torch
+
+from sparkagd_amd import (
+    LogisticGradient,
+    LeastSquaresGradient,
+    L1Updater,
+    SimpleUpdater,
+    SquaredL2Updater,
+    run,
+)
+from sparkagd_amd.data import generate_dense_problem, generate_logistic_data
+from sparkagd_amd.gram import GramOperator, run_gram
+from sparkagd_amd import ops
+
+
+def _args(data, updater, reg, w0, iters=12, tol=1e-12):
+    return (data, LogisticGradient(), updater, tol, iters, reg, w0,
+            1.0, math.inf, 0.5, 0.9, True)
+
+
+@pytest.mark.parametrize("reg,updater_cls", [(0.0, SimpleUpdater), (0.2, SquaredL2Updater)])
+def test_gram_matches_direct(reg, updater_cls):
+    data = generate_logistic_data(2.0, -1.5, 4000, seed=21)
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+    w_d, h_d = run(*_args(data, updater_cls(), reg, w0))
+    w_g, h_g = run(*_args(data, updater_cls(), reg, w0), solver="gram")
+    assert len(h_d) == len(h_g)
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 1e-8 * max(1.0, abs(b)), (a, b)
+    torch.testing.assert_close(w_g, w_d, rtol=1e-7, atol=1e-9)
+
+
+def test_gram_nonzero_x0_and_wide_d():
+    shard, _ = generate_dense_problem(300, 900, seed=4, dtype=torch.float64)
+    w0 = torch.randn(900, dtype=torch.float64, generator=torch.Generator().manual_seed(0)) * 0.1
+    args = (shard, LogisticGradient(), SquaredL2Updater(), 1e-12, 10, 0.05, w0,
+            1.0, math.inf, 0.5, 0.9, True)
+    w_d, h_d = run(*args)
+    w_g, h_g = run(*args, solver="gram")
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 1e-8 * max(1.0, abs(b))
+    torch.testing.assert_close(w_g, w_d, rtol=1e-6, atol=1e-8)
+
+
+def test_gram_least_squares_beta_ge_1():
+    shard, _ = generate_dense_problem(200, 500, seed=7,
+                                      loss_type=ops.LOSS_LEAST_SQUARES,
+                                      dtype=torch.float64)
+    w0 = torch.zeros(500, dtype=torch.float64)
+    feats = shard.features
+    Lsafe = float(2 * (feats * feats).sum() / feats.shape[0])
+    args = (shard, LeastSquaresGradient(), SimpleUpdater(), 1e-12, 15, 0.0, w0,
+            Lsafe, Lsafe, 1.5, 1.0, True)
+    w_d, h_d = run(*args)
+    w_g, h_g = run(*args, solver="gram", loss_history_mode="backtrack")
+    assert len(h_d) == len(h_g)
+    torch.testing.assert_close(w_g, w_d, rtol=1e-6, atol=1e-8)
+
+
+def test_gram_rejects_nonaffine_and_checkpoint():
+    data = generate_logistic_data(2.0, -1.5, 500, seed=1)
+    w0 = torch.zeros(2, dtype=torch.float64)
+    with pytest.raises(ValueError):
+        run(*_args(data, L1Updater(), 0.1, w0), solver="gram")
+    with pytest.raises(ValueError):
+        run(*_args(data, SimpleUpdater(), 0.0, w0), solver="gram",
+            checkpoint_path="/tmp/x.safetensors", checkpoint_every=1)
+
+
+def test_gram_operator_reuse():
+    """A prebuilt GramOperator can be reused across solves (warm restarts)."""
+    from sparkagd_amd.parallel.comm import Communicator
+
+    data = generate_logistic_data(2.0, -1.5, 1000, seed=3)
+    op = GramOperator(data, Communicator())
+    assert op.K.shape == (1000, 1000)
+    w0 = torch.zeros(2, dtype=torch.float64)
+    w1, h1 = run_gram(data, LogisticGradient(), SimpleUpdater(), 1e-12, 5, 0.0,
+                      w0, 1.0, math.inf, 0.5, 0.9, True, gram_op=op)
+    w2, h2 = run_gram(data, LogisticGradient(), SimpleUpdater(), 1e-12, 5, 0.0,
+                      w0, 1.0, math.inf, 0.5, 0.9, True, gram_op=op)
+    assert torch.equal(w1, w2) and h1 == h2
