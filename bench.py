@@ -75,6 +75,14 @@ class CountingGradient:
         self.n_passes += 1 if need_grad else 0
         return self.inner.eval_from_margins(shard, margins, mask, need_grad)
 
+    def multiplier_loss(self, shard, margins, mask=None):
+        self.n_evals += 1  # a loss evaluation over every example (n-space)
+        return self.inner.multiplier_loss(shard, margins, mask)
+
+    @property
+    def LOSS_TYPE(self):
+        return self.inner.LOSS_TYPE
+
 
 def sync(device: torch.device) -> None:
     if device.type == "cuda":

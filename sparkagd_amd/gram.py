@@ -146,13 +146,12 @@ def run_gram(
     if getattr(data, "kind", None) != "dense":
         raise ValueError("run_gram requires a DenseShard")
     backtrack_tol = 1e-10
-    loss_type = gradient.LOSS_TYPE
 
     op = gram_op or GramOperator(data, comm)
     acc = op.acc
     dev = data.device
     n_local = data.n
-    feats, labels = data.features, data.labels
+    feats = data.features
 
     # count c (full batch, constant)
     cvec = torch.tensor([float(n_local)], dtype=torch.float64, device=dev)
@@ -161,7 +160,7 @@ def run_gram(
 
     x0 = initial_weights.clone()
     x0_nonzero = bool(torch.any(x0 != 0))
-    xm0 = data.margins(x0.to(acc)) if x0_nonzero else torch.zeros(n_local, dtype=acc, device=dev)
+    xm0 = gradient.margins(data, x0.to(acc)) if x0_nonzero else torch.zeros(n_local, dtype=acc, device=dev)
     norm_x0_sq = float((x0.to(torch.float64) ** 2).sum())
 
     # basis storage (index 0 = x0; gradients at 1..T)
@@ -205,7 +204,7 @@ def run_gram(
 
     def eval_loss(vm: torch.Tensor) -> Tuple[float, torch.Tensor]:
         """(mean loss, multiplier) at tracked margins vm — zero data passes."""
-        mult, lc = ops.dense_multiplier_loss(feats, vm, labels, loss_type)
+        mult, lc = gradient.multiplier_loss(data, vm)
         comm.allreduce_(lc)
         return float(lc[0]) / c, mult
 

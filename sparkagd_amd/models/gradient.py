@@ -54,6 +54,13 @@ class Gradient:
         """Loss (+ gradient) from precomputed margins — saves the A·w pass."""
         return shard.eval_from_margins(margins, self.LOSS_TYPE, mask, need_grad)
 
+    def multiplier_loss(self, shard, margins: torch.Tensor,
+                        mask: Optional[torch.Tensor] = None):
+        """(multiplier, loss_count) from precomputed margins (zero data
+        passes) — the Gram solver's n-space evaluation."""
+        return ops.dense_multiplier_loss(shard.features, margins, shard.labels,
+                                         self.LOSS_TYPE, mask)
+
     # --- MLlib per-example API parity (reference Gradient.compute) ---
     def compute(
         self,
