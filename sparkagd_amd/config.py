@@ -56,6 +56,10 @@ class AGDConfig:
     alpha: float = 0.9
     may_restart: bool = True
     loss_history_mode: str = "exact"
+    #: 'direct' streams the shard (2 passes/iteration with margin tracking);
+    #: 'gram' precomputes K = A·A^T and iterates in O(n_local·n_global)
+    #: (dense shards + affine prox only; see sparkagd_amd/gram.py).
+    solver: str = "direct"
 
     # Numerical guard below which the simple backtracking test switches to the
     # alternate (cancellation-safe) test (reference ``AGD.scala:234-235,272-278``).
@@ -85,3 +89,5 @@ class AGDConfig:
             raise ValueError("beta must be > 0")
         if self.loss_history_mode not in ("exact", "backtrack", "none"):
             raise ValueError("loss_history_mode must be exact|backtrack|none")
+        if self.solver not in ("direct", "gram"):
+            raise ValueError("solver must be direct|gram")

@@ -565,5 +565,6 @@ class AcceleratedGradientDescent(Optimizer):
             checkpoint_path=self.checkpoint_path,
             checkpoint_every=self.checkpoint_every,
             resume_from=self.resume_from,
+            solver=c.solver,
         )
         return weights

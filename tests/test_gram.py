@@ -62,6 +62,30 @@ def test_gram_least_squares_beta_ge_1():
     torch.testing.assert_close(w_g, w_d, rtol=1e-6, atol=1e-8)
 
 
+def test_gram_hinge_converges():
+    """Hinge multiplier is discontinuous, so bitwise trajectory identity is
+    not expected; the Gram solver must still converge."""
+    from sparkagd_amd import HingeGradient
+
+    shard, _ = generate_dense_problem(1000, 300, seed=9,
+                                      loss_type=ops.LOSS_HINGE,
+                                      dtype=torch.float64)
+    w0 = torch.zeros(300, dtype=torch.float64)
+    w, h = run(shard, HingeGradient(), SquaredL2Updater(), 1e-12, 20, 0.01, w0,
+               1.0, math.inf, 0.5, 0.9, True, solver="gram")
+    assert h[-1] < h[0] and h[-1] < 1.0
+
+
+def test_gram_via_class_api_config():
+    from sparkagd_amd import AcceleratedGradientDescent, AGDConfig
+
+    data = generate_logistic_data(2.0, -1.5, 1500, seed=2)
+    cfg = AGDConfig(num_iterations=6, convergence_tol=1e-12, solver="gram")
+    opt = AcceleratedGradientDescent(LogisticGradient(), SimpleUpdater(), cfg)
+    w = opt.optimize(data, torch.zeros(2, dtype=torch.float64))
+    assert len(opt.loss_history) == 6
+
+
 def test_gram_rejects_nonaffine_and_checkpoint():
     data = generate_logistic_data(2.0, -1.5, 500, seed=1)
     w0 = torch.zeros(2, dtype=torch.float64)
