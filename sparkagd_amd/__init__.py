@@ -31,6 +31,7 @@ from .models.updater import (
 )
 from .data import DenseShard, CSRShard, generate_logistic_data, generate_dense_problem
 from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniBatch
+from .gram import GramOperator, run_gram
 from .parallel.comm import Communicator
 
 __version__ = "0.1.0"
@@ -54,4 +55,6 @@ __all__ = [
     "run_mini_batch",
     "runMiniBatch",
     "Communicator",
+    "GramOperator",
+    "run_gram",
 ]

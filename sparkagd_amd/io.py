@@ -10,9 +10,8 @@ device-resident shards from host arrays, scipy CSR matrices, svmlight files
 
 from __future__ import annotations
 
-import json
 import os
-from typing import Optional, Tuple, Union
+from typing import Optional, Union
 
 import numpy as np
 import torch

@@ -14,7 +14,6 @@ from typing import Optional
 
 import torch
 
-from .. import ops
 from ..config import AGDConfig
 from ..optimizer import AcceleratedGradientDescent
 from ..parallel.comm import Communicator
