@@ -22,6 +22,7 @@ from .models.gradient import (
     LogisticGradient,
     LeastSquaresGradient,
     HingeGradient,
+    SmoothedHingeGradient,
 )
 from .models.updater import (
     Updater,
@@ -42,6 +43,7 @@ __all__ = [
     "LogisticGradient",
     "LeastSquaresGradient",
     "HingeGradient",
+    "SmoothedHingeGradient",
     "Updater",
     "SimpleUpdater",
     "L1Updater",

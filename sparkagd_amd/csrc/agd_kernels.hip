@@ -19,11 +19,11 @@
 //    lane, one wave per row (dense A·w) / contiguous 256-thread column slabs
 //    (dense A^T·m), grid-stride everywhere, fp32 accumulation (fp64 for fp64
 //    shards), fp64 block reductions for the loss and the iteration scalars.
-//  * A^T·m is deterministic by construction: each (row-block, column-slab)
-//    workgroup writes a private partial slab; a second kernel reduces the
-//    row-block axis. No global atomics on the dense path => bitwise
-//    reproducible gradients (the CSR A^T·m uses fp32 atomics; see
-//    csr_grad_deterministic note in ops/hiplib.py).
+//  * Deterministic by construction: dense A^T·m writes private partial
+//    slabs reduced in fixed order; CSR A^T·m gathers over a CSC copy by
+//    default (the fp32 atomic scatter remains an option); all fp64 scalar
+//    reductions are two-stage (block partials + one-block final) — no
+//    same-word atomics anywhere on the default paths.
 //  * No Triton, no CUDA-compat headers, no torch headers: plain HIP + a C ABI
 //    taking raw device pointers and a hipStream_t, loaded via ctypes.
 //
@@ -396,6 +396,7 @@ __global__ __launch_bounds__(BLOCK) void k_slab_reduce(TACC* __restrict__ part,
 #define LOSS_LOGISTIC 0
 #define LOSS_LSQ 1
 #define LOSS_HINGE 2
+#define LOSS_SMOOTH_HINGE 3  // quadratically smoothed (Rennie) hinge
 
 template <typename TACC>
 __device__ __forceinline__ TACC softplus(TACC t) {
@@ -432,11 +433,24 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
       const TACC diff = z - y;
       m = (TACC)2 * diff;
       l = diff * diff;
-    } else {
+    } else if (loss_type == LOSS_HINGE) {
       const TACC s = (TACC)2 * y - (TACC)1;
       const TACC sz = s * z;
       m = (sz < (TACC)1) ? -s : (TACC)0;
       l = (sz < (TACC)1) ? (TACC)1 - sz : (TACC)0;
+    } else {  // LOSS_SMOOTH_HINGE: 0 if sz>=1; (1-sz)^2/2 if 0<sz<1; 0.5-sz else
+      const TACC s = (TACC)2 * y - (TACC)1;
+      const TACC sz = s * z;
+      if (sz >= (TACC)1) {
+        m = (TACC)0;
+        l = (TACC)0;
+      } else if (sz > (TACC)0) {
+        m = -s * ((TACC)1 - sz);
+        l = (TACC)0.5 * ((TACC)1 - sz) * ((TACC)1 - sz);
+      } else {
+        m = -s;
+        l = (TACC)0.5 - sz;
+      }
     }
     mult[i] = m;
     lsum += (double)l;

@@ -7,7 +7,7 @@ surface is what the reference's tests exercise — but the implementations are
 batched GPU kernels, not per-example JVM loops.
 """
 
-from .gradient import Gradient, LogisticGradient, LeastSquaresGradient, HingeGradient
+from .gradient import Gradient, LogisticGradient, LeastSquaresGradient, HingeGradient, SmoothedHingeGradient
 from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater
 from .trainers import LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD
 
@@ -16,6 +16,7 @@ __all__ = [
     "LogisticGradient",
     "LeastSquaresGradient",
     "HingeGradient",
+    "SmoothedHingeGradient",
     "Updater",
     "SimpleUpdater",
     "L1Updater",

@@ -104,6 +104,18 @@ class HingeGradient(Gradient):
     """Hinge loss (linear SVM); labels in {0, 1}, scaled to s = 2y-1 in {-1, 1}.
 
     loss_i = max(0, 1 - s z), mult_i = -s when s z < 1 else 0.
+    NOTE: the hinge is nonsmooth; the accelerated method's backtracking
+    assumes a Lipschitz gradient, so SmoothedHingeGradient usually converges
+    much better under AGD. Plain hinge is kept for MLlib parity.
     """
 
     LOSS_TYPE = ops.LOSS_HINGE
+
+
+class SmoothedHingeGradient(Gradient):
+    """Rennie's quadratically smoothed hinge (differentiable SVM loss):
+    loss = 0 if sz>=1; (1-sz)^2/2 if 0<sz<1; 0.5-sz otherwise. Capability
+    beyond the reference: a hinge-family loss that satisfies the smoothness
+    assumptions of the accelerated method."""
+
+    LOSS_TYPE = ops.LOSS_SMOOTH_HINGE

@@ -25,6 +25,7 @@ from .reference import (
     LOSS_LOGISTIC,
     LOSS_LEAST_SQUARES,
     LOSS_HINGE,
+    LOSS_SMOOTH_HINGE,
     PROX_SIMPLE,
     PROX_L1,
     PROX_SQUARED_L2,
@@ -177,6 +178,7 @@ __all__ = [
     "LOSS_LOGISTIC",
     "LOSS_LEAST_SQUARES",
     "LOSS_HINGE",
+    "LOSS_SMOOTH_HINGE",
     "PROX_SIMPLE",
     "PROX_L1",
     "PROX_SQUARED_L2",

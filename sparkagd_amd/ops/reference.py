@@ -30,6 +30,7 @@ import torch
 LOSS_LOGISTIC = 0
 LOSS_LEAST_SQUARES = 1
 LOSS_HINGE = 2
+LOSS_SMOOTH_HINGE = 3
 
 PROX_SIMPLE = 0
 PROX_L1 = 1
@@ -54,6 +55,16 @@ def _multiplier_and_loss(
         viol = s * z < 1.0
         mult = torch.where(viol, -s, torch.zeros_like(z))
         loss = torch.clamp(1.0 - s * z, min=0.0)
+    elif loss_type == LOSS_SMOOTH_HINGE:
+        # Rennie's quadratically smoothed hinge: differentiable, so the
+        # accelerated method's smoothness assumptions hold (plain hinge is
+        # nonsmooth and is kept for MLlib parity).
+        s = 2.0 * y - 1.0
+        sz = s * z
+        mult = torch.where(sz >= 1.0, torch.zeros_like(z),
+                           torch.where(sz > 0.0, -s * (1.0 - sz), -s))
+        loss = torch.where(sz >= 1.0, torch.zeros_like(z),
+                           torch.where(sz > 0.0, 0.5 * (1.0 - sz) ** 2, 0.5 - sz))
     else:
         raise ValueError(f"unknown loss_type {loss_type}")
     return mult, loss

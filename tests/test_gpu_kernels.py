@@ -23,7 +23,7 @@ def _mk_dense(n, d, dtype, seed=0):
 
 
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
-@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES, ops.LOSS_HINGE])
+@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES, ops.LOSS_HINGE, ops.LOSS_SMOOTH_HINGE])
 def test_dense_eval_matches_reference(dtype, loss_type):
     from sparkagd_amd.ops import hiplib, reference
 

@@ -29,15 +29,20 @@ def _np_eval(A, y, w, loss_type):
     elif loss_type == ops.LOSS_LEAST_SQUARES:
         mult = 2.0 * (z - y)
         loss = (z - y) ** 2
-    else:
+    elif loss_type == ops.LOSS_HINGE:
         s = 2.0 * y - 1.0
         viol = s * z < 1.0
         mult = np.where(viol, -s, 0.0)
         loss = np.maximum(0.0, 1.0 - s * z)
+    else:  # smoothed hinge
+        s = 2.0 * y - 1.0
+        sz = s * z
+        mult = np.where(sz >= 1.0, 0.0, np.where(sz > 0.0, -s * (1.0 - sz), -s))
+        loss = np.where(sz >= 1.0, 0.0, np.where(sz > 0.0, 0.5 * (1.0 - sz) ** 2, 0.5 - sz))
     return A.T @ mult, loss.sum()
 
 
-@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES, ops.LOSS_HINGE])
+@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES, ops.LOSS_HINGE, ops.LOSS_SMOOTH_HINGE])
 def test_dense_eval_matches_numpy(loss_type):
     rng = np.random.default_rng(0)
     n, d = 257, 13

@@ -127,6 +127,27 @@ def test_no_backtracking_beta_ge_1(data):
     assert hist[-1] < hist[0]
 
 
+def test_smoothed_hinge_beats_plain_hinge_under_agd():
+    """AGD assumes a Lipschitz gradient; the smoothed hinge satisfies it and
+    should reach a separating solution where the plain hinge stalls."""
+    from sparkagd_amd.data import generate_dense_problem
+    from sparkagd_amd.models.gradient import HingeGradient, SmoothedHingeGradient
+    from sparkagd_amd import ops as _ops
+
+    shard, _ = generate_dense_problem(3000, 50, seed=17,
+                                      loss_type=_ops.LOSS_HINGE,
+                                      dtype=torch.float64)
+    w0 = torch.zeros(50, dtype=torch.float64)
+    args = (1e-10, 25, 0.001, w0, 1.0, math.inf, 0.5, 0.9, True)
+    _, h_s = run(shard, SmoothedHingeGradient(), SquaredL2Updater(), *args)
+    assert h_s[-1] < 0.25 * h_s[0]  # converges well
+    # classification accuracy with the smoothed-hinge solution
+    w_s, _ = run(shard, SmoothedHingeGradient(), SquaredL2Updater(), *args)
+    pred = (shard.features @ w_s > 0).double()
+    acc = float((pred == shard.labels).double().mean())
+    assert acc > 0.9
+
+
 def test_nan_guard():
     """NaN loss -> warn + clean break (AGD.scala:309-312)."""
     feats = torch.tensor([[1e200, 1e200]], dtype=torch.float64)
