@@ -30,6 +30,7 @@
 // Build: hipcc --offload-arch=gfx950 -O3 -shared -fPIC agd_kernels.hip -o libagd_hip.so
 
 #include <hip/hip_runtime.h>
+#include <hipblaslt/hipblaslt.h>
 
 #include <cmath>
 #include <cstdint>
@@ -999,6 +1000,75 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
     hipLaunchKernelGGL(k_csr_grad, dim3(grid), dim3(BLOCK), 0, s, rp, ci, v,
                        mult, n, (float*)grad_out);
   }
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Library GEMM for the Gram-operator build (the one true GEMM in this
+// workload): C[m,n] (f32, row-major) = A[m,k] (bf16 rm) · B[n,k]^T (bf16 rm),
+// fp32 accumulation via hipBLASLt (bf16 in / f32 out — not expressible
+// through torch.matmul, which rounds the output to bf16). Row-major is
+// mapped to hipBLASLt's column-major as C'[n,m] = B'^T(k x n) · A'(k x m).
+// ---------------------------------------------------------------------------
+
+#define LT_CHECK(expr)                                                         \
+  do {                                                                         \
+    hipblasStatus_t _s = (expr);                                               \
+    if (_s != HIPBLAS_STATUS_SUCCESS) {                                        \
+      snprintf(g_err, sizeof(g_err), "%s:%d %s: hipblaslt status %d",          \
+               __FILE__, __LINE__, #expr, (int)_s);                            \
+      return 3;                                                                \
+    }                                                                          \
+  } while (0)
+
+static hipblasLtHandle_t g_lt_handle = nullptr;
+static void* g_lt_ws = nullptr;
+static const size_t LT_WS_BYTES = 64u << 20;
+
+extern "C" int agd_gemm_bf16f32_nt(const void* A, const void* B, void* C,
+                                   long long m, long long n, long long k,
+                                   float beta, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  if (g_lt_handle == nullptr) {
+    LT_CHECK(hipblasLtCreate(&g_lt_handle));
+    HIP_CHECK(hipMalloc(&g_lt_ws, LT_WS_BYTES));
+  }
+  hipblasLtMatmulDesc_t op = nullptr;
+  hipblasLtMatrixLayout_t la = nullptr, lb = nullptr, lc = nullptr;
+  LT_CHECK(hipblasLtMatmulDescCreate(&op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  const hipblasOperation_t opT = HIPBLAS_OP_T, opN = HIPBLAS_OP_N;
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSA,
+                                           &opT, sizeof(opT)));
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSB,
+                                           &opN, sizeof(opN)));
+  // col-major views: "A" = B' (k x n, ld k) transposed; "B" = A' (k x m, ld k)
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, k, n, k));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, k, m, k));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&lc, HIP_R_32F, n, m, n));
+  const float alpha = 1.0f;
+
+  hipblasLtMatmulPreference_t pref = nullptr;
+  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &LT_WS_BYTES,
+      sizeof(LT_WS_BYTES)));
+  hipblasLtMatmulHeuristicResult_t heur;
+  int found = 0;
+  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(g_lt_handle, op, la, lb, lc, lc,
+                                           pref, 1, &heur, &found));
+  if (found < 1) {
+    snprintf(g_err, sizeof(g_err), "hipblaslt: no algo for bf16f32 %lldx%lldx%lld",
+             m, n, k);
+    return 3;
+  }
+  LT_CHECK(hipblasLtMatmul(g_lt_handle, op, &alpha, B, la, A, lb, &beta, C, lc,
+                           C, lc, &heur.algo, g_lt_ws, LT_WS_BYTES, s));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  hipblasLtMatrixLayoutDestroy(lc);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatmulDescDestroy(op);
   HIP_CHECK(hipGetLastError());
   return 0;
 }

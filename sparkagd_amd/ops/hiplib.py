@@ -66,6 +66,8 @@ def load() -> ctypes.CDLL:
     lib.agd_fused_scalars.argtypes = [P, P, P, P, P, LL, I, P, P]
     lib.agd_dot_diff.restype = I
     lib.agd_dot_diff.argtypes = [P, P, P, P, P, LL, I, P, P]
+    lib.agd_gemm_bf16f32_nt.restype = I
+    lib.agd_gemm_bf16f32_nt.argtypes = [P, P, P, LL, LL, LL, ctypes.c_float, P]
 
     _lib = lib
     return lib
@@ -323,6 +325,23 @@ def csr_eval(
     )
     _check(rc)
     return grad, loss_count
+
+
+def gemm_bf16f32_nt(A: torch.Tensor, B: torch.Tensor, C: torch.Tensor,
+                    beta: float = 0.0) -> torch.Tensor:
+    """C[m,n] (f32) = A[m,k] (bf16) @ B[n,k]^T (bf16) + beta*C via hipBLASLt
+    (fp32 accumulation, fp32 output — used by the Gram-operator build)."""
+    lib = load()
+    assert A.dtype == torch.bfloat16 and B.dtype == torch.bfloat16
+    assert C.dtype == torch.float32
+    m, k = A.shape
+    n, k2 = B.shape
+    assert k == k2 and C.shape == (m, n)
+    assert A.is_contiguous() and B.is_contiguous() and C.is_contiguous()
+    rc = lib.agd_gemm_bf16f32_nt(_ptr(A), _ptr(B), _ptr(C), m, n, k,
+                                 float(beta), _stream(A))
+    _check(rc)
+    return C
 
 
 def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
