@@ -193,8 +193,10 @@ def run_gram(
     xm0_64 = xm0.to(torch.float64)
     T = 0  # gradient basis vectors so far
 
-    def new_basis(m_t: torch.Tensor) -> Tuple[int, torch.Tensor]:
-        """Register gradient basis vector v = Aᵀ m_t / c; returns (index, gm)."""
+    def new_basis_async(m_t: torch.Tensor) -> Tuple[int, torch.Tensor, torch.Tensor]:
+        """Enqueue registration of gradient basis vector v = Aᵀ m_t / c:
+        returns (index, gm, row_device) with the G-row still on device —
+        call finish_basis(index, row_device) before using G entries for it."""
         nonlocal T
         if T >= max_basis:
             raise RuntimeError("gram basis overflow — raise max_basis")
@@ -210,23 +212,33 @@ def run_gram(
             row[1: t] = GMstore[:T] @ md
         row[t] = (md * gm.to(torch.float64)).sum()
         comm.allreduce_(row)
-        row_h = (row / c).cpu().numpy()
-        G[t, : t + 1] = row_h
-        G[: t + 1, t] = row_h
         Mstore[T] = m_t
         GMstore[T] = gm.to(torch.float64)
         GMstore32[T] = gm
         T += 1
-        return t, gm
+        return t, gm, row
+
+    def finish_basis(t: int, row: torch.Tensor) -> None:
+        row_h = (row / c).cpu().numpy()
+        G[t, : t + 1] = row_h
+        G[: t + 1, t] = row_h
 
     def quad(a: np.ndarray, b: np.ndarray, k: int) -> float:
         return float(a[:k] @ (G[:k, :k] @ b[:k]))
 
-    def eval_loss(vm: torch.Tensor) -> Tuple[float, torch.Tensor]:
-        """(mean loss, multiplier) at tracked margins vm — zero data passes."""
+    def eval_loss_async(vm: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Enqueue multiplier/loss at tracked margins vm (zero data passes);
+        returns (loss_count_device, multiplier)."""
         mult, lc = gradient.multiplier_loss(data, vm)
         comm.allreduce_(lc)
-        return float(lc[0]) / c, mult
+        return lc, mult
+
+    def read_loss(lc: torch.Tensor) -> float:
+        return float(lc[0]) / c
+
+    def eval_loss(vm: torch.Tensor) -> Tuple[float, torch.Tensor]:
+        lc, mult = eval_loss_async(vm)
+        return read_loss(lc), mult
 
     def coef(vec_len: int) -> np.ndarray:
         return np.zeros(vec_len)
@@ -273,8 +285,12 @@ def run_gram(
             theta = 2.0 / (1.0 + math.sqrt(1.0 + 4.0 * (L / L_old) / (theta_old * theta_old)))
             cy = (1.0 - theta) * cx_old + theta * cz_old
             ym = ops.axpby(1.0 - theta, xm_old, theta, zm_old)
-            f_y, m_y = eval_loss(ym)
-            t_y, gm_y = new_basis(m_y)
+            # Enqueue the whole trial's device work before the first host
+            # read: y-loss, basis registration (K·m + G row), the margin
+            # updates, and (simple mode) the x-loss — then pay ONE pipeline
+            # wait instead of three.
+            lc_y, m_y = eval_loss_async(ym)
+            t_y, gm_y, row_y = new_basis_async(m_y)
             step = 1.0 / (theta * L)
             cz = prox_coeff(cz_old, t_y, step)
             cx = (1.0 - theta) * cx_old + theta * cz
@@ -282,7 +298,15 @@ def run_gram(
             xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
 
             if beta >= 1.0:
+                f_y = read_loss(lc_y)
+                finish_basis(t_y, row_y)
                 break
+
+            lc_x = None
+            if backtrack_simple:
+                lc_x, _ = eval_loss_async(xm)
+            f_y = read_loss(lc_y)
+            finish_basis(t_y, row_y)
 
             k = T + 1
             dxy = cx - cy
@@ -291,7 +315,10 @@ def run_gram(
                 break
 
             if backtrack_simple:
-                f_x, _ = eval_loss(xm)
+                # x == y is excluded above, so reading the pre-enqueued f_x
+                # here matches the reference's compute-f_x-after-the-check
+                # order (and at xy_sq == 0 its value would equal f_y anyway).
+                f_x = read_loss(lc_x)
                 f_x_bt = f_x
                 xy_dot_gy = float(dxy[:k] @ G[:k, t_y])
                 q_x = f_y + xy_dot_gy + 0.5 * L * xy_sq
@@ -300,7 +327,8 @@ def run_gram(
             else:
                 f_x, m_x = eval_loss(xm)
                 f_x_bt = f_x
-                t_x, _gm_x = new_basis(m_x)
+                t_x, _gm_x, row_x = new_basis_async(m_x)
+                finish_basis(t_x, row_x)
                 k = T + 1
                 localL = 2.0 * float(dxy[:k] @ (G[:k, t_x] - G[:k, t_y])) / xy_sq
 
