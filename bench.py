@@ -96,7 +96,7 @@ def main() -> int:
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--rows", type=int, default=16384, help="rows per GPU (weak scaling)")
     p.add_argument("--d", type=int, default=1_000_000)
-    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "f32"])
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "f32", "f8"])
     p.add_argument("--loss", type=str, default="logistic", choices=list(LOSSES))
     p.add_argument("--reg", type=float, default=0.0)
     p.add_argument("--csr", action="store_true", help="CSR-sparse shard instead of dense")
@@ -110,7 +110,8 @@ def main() -> int:
     rank, world = comm.rank, comm.world_size
     if torch.cuda.is_available():
         device = torch.device("cuda", torch.cuda.current_device())
-        dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+        dtype = {"bf16": torch.bfloat16, "f32": torch.float32,
+                 "f8": torch.float8_e4m3fn}[args.dtype]
         wdtype = torch.float32
     else:  # CPU smoke fallback (the real bench runs on MI355X)
         device = torch.device("cpu")

@@ -145,8 +145,23 @@ __global__ __launch_bounds__(BLOCK) void k_reduce_partials(
 template <typename TA, typename TACC, int W, bool NT = false>
 __device__ __forceinline__ void loadW(const TA* __restrict__ p, TACC (&out)[W]) {
   if constexpr (W == 1) {
-    if constexpr (sizeof(TA) == 2) out[0] = bf2f(*(const ubf16*)p);
+    if constexpr (sizeof(TA) == 1)
+      out[0] = __builtin_amdgcn_cvt_f32_fp8((int)*(const unsigned char*)p, 0);
+    else if constexpr (sizeof(TA) == 2) out[0] = bf2f(*(const ubf16*)p);
     else out[0] = (TACC)p[0];
+  } else if constexpr (sizeof(TA) == 1) {  // fp8 e4m3fn, W == 16 (16 B)
+    using i32x4 = __attribute__((ext_vector_type(4))) int;
+    using f32x2 = __attribute__((ext_vector_type(2))) float;
+    i32x4 v = NT ? __builtin_nontemporal_load((const i32x4*)p) : *(const i32x4*)p;
+#pragma unroll
+    for (int ch = 0; ch < 4; ++ch) {
+      const f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(v[ch], false);
+      const f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(v[ch], true);
+      out[ch * 4 + 0] = lo[0];
+      out[ch * 4 + 1] = lo[1];
+      out[ch * 4 + 2] = hi[0];
+      out[ch * 4 + 3] = hi[1];
+    }
   } else if constexpr (sizeof(TA) == 2) {  // bf16, W == 8 (16 B)
     using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
     u16x8 v = NT ? __builtin_nontemporal_load((const u16x8*)p) : *(const u16x8*)p;
@@ -777,7 +792,7 @@ static inline int grid_for(ll work_items, ll per_block) {
 // W (elements per 16-B lane load) for a features dtype, or 1 if the row
 // stride is not 16-B aligned.
 static inline int pick_w(int dtype, ll d) {
-  const int wfull = (dtype == 0) ? 8 : (dtype == 1) ? 4 : 2;
+  const int wfull = (dtype == 3) ? 16 : (dtype == 0) ? 8 : (dtype == 1) ? 4 : 2;
   return (d % wfull == 0) ? wfull : 1;
 }
 
@@ -944,6 +959,8 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
     case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
     case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
     case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 31: return dense_eval_t<unsigned char, float, 16>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 30: return dense_eval_t<unsigned char, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;

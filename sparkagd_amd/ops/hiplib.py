@@ -25,8 +25,8 @@ from typing import Optional, Tuple
 
 import torch
 
-_DTYPE_CODE = {torch.bfloat16: 0, torch.float32: 1, torch.float64: 2}
-_ACC_DTYPE = {torch.bfloat16: torch.float32, torch.float32: torch.float32, torch.float64: torch.float64}
+_DTYPE_CODE = {torch.bfloat16: 0, torch.float32: 1, torch.float64: 2, torch.float8_e4m3fn: 3}
+_ACC_DTYPE = {torch.bfloat16: torch.float32, torch.float32: torch.float32, torch.float64: torch.float64, torch.float8_e4m3fn: torch.float32}
 
 _lib: Optional[ctypes.CDLL] = None
 

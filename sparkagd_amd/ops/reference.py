@@ -87,9 +87,9 @@ def dense_eval(
     Returns ``(grad_sum, loss_count)`` where grad_sum has w's dtype and shape
     [d], and loss_count is float64 [2] = (sum of losses, number of examples).
     """
-    acc_dtype = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    acc_dtype = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
     wa = w.to(acc_dtype)
-    z = (features.to(acc_dtype) @ wa) if features.dtype in (torch.bfloat16, torch.float16) else (features @ wa)
+    z = (features.to(acc_dtype) @ wa) if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else (features @ wa)
     mult, loss = _multiplier_and_loss(z, labels, loss_type)
     if mask is not None:
         m = mask.to(mult.dtype)
@@ -141,7 +141,7 @@ def csr_eval(
 
 def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
     """margins = A @ v (margin-state tracking support)."""
-    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
     return features.to(acc) @ v.to(acc)
 
 
@@ -153,7 +153,7 @@ def dense_eval_from_margins(
     mask: Optional[torch.Tensor] = None,
     need_grad: bool = True,
 ) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
-    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
     mult, loss = _multiplier_and_loss(margins, labels, loss_type)
     if mask is not None:
         m = mask.to(mult.dtype)
@@ -188,7 +188,7 @@ def dense_multiplier_loss(
 
 
 def dense_grad_from_mult(features: torch.Tensor, mult: torch.Tensor) -> torch.Tensor:
-    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
     return features.to(acc).T @ mult.to(acc)
 
 

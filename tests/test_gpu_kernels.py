@@ -134,6 +134,27 @@ def test_margins_mfma_matches_valu(n, d, monkeypatch):
     assert float(lc1[1]) == float(lc2[1]) == n
 
 
+@pytest.mark.parametrize("loss_type", [ops.LOSS_LOGISTIC, ops.LOSS_LEAST_SQUARES])
+def test_dense_eval_fp8(loss_type):
+    """fp8 e4m3fn shard path (hardware v_cvt_pk_f32_fp8 decodes) vs the
+    oracle evaluated on the same quantized values."""
+    from sparkagd_amd.ops import hiplib, reference
+
+    A32, y, w = _mk_dense(4096, 512, torch.float32, seed=41)
+    A8 = A32.to(torch.float8_e4m3fn).contiguous()
+    grad_h, lc_h = hiplib.dense_eval(A8, y, w, loss_type)
+    grad_r, lc_r = reference.dense_eval(A8, y, w, loss_type)
+    torch.testing.assert_close(grad_h, grad_r, rtol=2e-4, atol=2e-3)
+    torch.testing.assert_close(lc_h, lc_r, rtol=1e-5, atol=1e-5)
+    # loss-only and unaligned-d scalar decode path
+    _, lc2 = hiplib.dense_eval(A8, y, w, loss_type, need_grad=False)
+    assert torch.equal(lc_h, lc2)
+    A8u = A8[:, :500].contiguous()
+    g_u, lc_u = hiplib.dense_eval(A8u, y, w[:500].contiguous(), loss_type)
+    g_ur, lc_ur = reference.dense_eval(A8u, y, w[:500], loss_type)
+    torch.testing.assert_close(g_u, g_ur, rtol=2e-4, atol=2e-3)
+
+
 def test_dense_eval_loss_only():
     """need_grad=False returns the identical loss/count without the A^T·m pass."""
     from sparkagd_amd.ops import hiplib

@@ -170,6 +170,20 @@ def test_axpby():
     torch.testing.assert_close(out, 0.3 * x - 1.7 * y)
 
 
+def test_fp8_reference_eval():
+    """float8_e4m3fn shards evaluate through the oracle (f32 accumulation)."""
+    from sparkagd_amd.data import generate_dense_problem
+
+    shard, _ = generate_dense_problem(128, 64, seed=5, dtype=torch.float8_e4m3fn)
+    w = torch.randn(64, dtype=torch.float32) * 0.1
+    grad, lc = shard.eval(w, ops.LOSS_LOGISTIC)
+    g_ref, lc_ref = ops.reference.dense_eval(
+        shard.features.to(torch.float32), shard.labels, w, ops.LOSS_LOGISTIC
+    )
+    torch.testing.assert_close(grad, g_ref, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(lc, lc_ref)
+
+
 def test_shard_properties():
     sh = DenseShard(torch.randn(10, 4), torch.zeros(10))
     assert sh.n == 10 and sh.d == 4 and sh.nbytes > 0
