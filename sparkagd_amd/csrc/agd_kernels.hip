@@ -244,13 +244,14 @@ __device__ __forceinline__ void storeAcc(TACC* __restrict__ p, const TACC (&v)[W
 // the slab axis => deterministic (no atomics anywhere on the dense path).
 // ---------------------------------------------------------------------------
 
-#define MARGIN_ROWS 4  // R: rows per wave (w-load amortization)
+#define MARGIN_ROWS 4   // R: rows per wave (w-load amortization)
+#define MARGIN_ROWS_FP8 8  // 1-byte elements: R=4 would make w-bytes ~ A-bytes
 
 template <typename TA, typename TACC, int W, bool NT>
 __global__ __launch_bounds__(BLOCK) void k_dense_margins(
     const TA* __restrict__ A, const TACC* __restrict__ w, ll n, ll d,
     ll slab_w, int n_slabs, TACC* __restrict__ part) {
-  constexpr int R = MARGIN_ROWS;
+  constexpr int R = (sizeof(TA) == 1) ? MARGIN_ROWS_FP8 : MARGIN_ROWS;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
@@ -887,7 +888,8 @@ static int dense_eval_t(const void* A, const float* labels,
                        stream, (const ubf16*)A, (const float*)w, n, d, slab_w,
                        n_slabs, (float*)margins);
   } else {
-    const ll tasks = ((n + MARGIN_ROWS - 1) / MARGIN_ROWS) * n_slabs;
+    const int R = (sizeof(TA) == 1) ? MARGIN_ROWS_FP8 : MARGIN_ROWS;
+    const ll tasks = ((n + R - 1) / R) * n_slabs;
     const int grid = grid_for(tasks, WAVES_PER_BLOCK);
     if (nt_loads)
       hipLaunchKernelGGL((k_dense_margins<TA, TACC, W, true>), dim3(grid),

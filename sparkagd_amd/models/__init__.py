@@ -9,7 +9,8 @@ batched GPU kernels, not per-example JVM loops.
 
 from .gradient import Gradient, LogisticGradient, LeastSquaresGradient, HingeGradient, SmoothedHingeGradient
 from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater
-from .trainers import LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD
+from .trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD,
+                       regularization_path)
 
 __all__ = [
     "Gradient",
@@ -24,4 +25,5 @@ __all__ = [
     "LogisticRegressionWithAGD",
     "LinearRegressionWithAGD",
     "SVMWithAGD",
+    "regularization_path",
 ]

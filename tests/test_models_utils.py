@@ -63,6 +63,26 @@ def test_minibatch_step_schedules():
                        1.0, w0, step_schedule="linear")
 
 
+def test_regularization_path_gram_reuse():
+    """One Gram build amortized over a lambda sweep; stronger regularization
+    shrinks the solution norm; results match per-lambda direct solves."""
+    from sparkagd_amd.models.trainers import regularization_path
+    from sparkagd_amd import LogisticGradient, SquaredL2Updater, run
+    import math as m
+
+    shard, _ = generate_dense_problem(600, 40, seed=6, dtype=torch.float64)
+    lambdas = [0.3, 0.03, 0.003]
+    models = regularization_path(shard, lambdas, num_iterations=20,
+                                 convergence_tol=1e-10, warm_start=False)
+    norms = [float(torch.norm(mod.weights)) for mod in models]
+    assert norms[0] < norms[1] < norms[2]
+    # cross-check the middle lambda against a direct solve
+    w_direct, _ = run(shard, LogisticGradient(), SquaredL2Updater(), 1e-10, 20,
+                      0.03, torch.zeros(40, dtype=torch.float64),
+                      1.0, m.inf, 0.5, 0.9, True)
+    torch.testing.assert_close(models[1].weights, w_direct, rtol=1e-6, atol=1e-8)
+
+
 def test_jsonl_metrics(tmp_path):
     p = str(tmp_path / "m.jsonl")
     with JsonlMetrics(p, rank=0) as m:
