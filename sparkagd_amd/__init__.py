@@ -29,6 +29,7 @@ from .models.updater import (
     SimpleUpdater,
     L1Updater,
     SquaredL2Updater,
+    ElasticNetUpdater,
 )
 from .data import DenseShard, CSRShard, generate_logistic_data, generate_dense_problem
 from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniBatch
@@ -48,6 +49,7 @@ __all__ = [
     "SimpleUpdater",
     "L1Updater",
     "SquaredL2Updater",
+    "ElasticNetUpdater",
     "DenseShard",
     "CSRShard",
     "generate_logistic_data",

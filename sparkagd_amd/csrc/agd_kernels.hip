@@ -661,10 +661,11 @@ __global__ __launch_bounds__(BLOCK) void k_axpby(double a, const T* __restrict__
 #define PROX_SIMPLE 0
 #define PROX_L1 1
 #define PROX_L2 2
+#define PROX_ELASTIC 3  // l1 soft-threshold then l2 shrink (elastic net)
 
 template <typename T>
 __device__ __forceinline__ T prox_elem(int kind, T wi, T gi, T ts, T tl,
-                                       double& racc) {
+                                       T tl2, double& racc) {
   if (kind == PROX_SIMPLE) return wi - ts * gi;
   if (kind == PROX_L1) {
     const T w1 = wi - ts * gi;
@@ -673,8 +674,19 @@ __device__ __forceinline__ T prox_elem(int kind, T wi, T gi, T ts, T tl,
     racc += (double)tl * fabs((double)wn);
     return wn;
   }
-  const T wn = wi * ((T)1 - ts * tl) - ts * gi;  // PROX_L2
-  racc += 0.5 * (double)tl * (double)wn * (double)wn;
+  if (kind == PROX_L2) {
+    const T wn = wi * ((T)1 - ts * tl) - ts * gi;
+    racc += 0.5 * (double)tl * (double)wn * (double)wn;
+    return wn;
+  }
+  // PROX_ELASTIC: prox of tl*|w|_1 + tl2/2*|w|^2 —
+  // soft-threshold by ts*tl, then shrink by 1/(1 + ts*tl2)
+  const T w1 = wi - ts * gi;
+  const T aw = fabs(w1) - tl * ts;
+  const T wsoft = (aw > (T)0) ? ((w1 > (T)0) ? aw : -aw) : (T)0;
+  const T wn = wsoft / ((T)1 + ts * tl2);
+  racc += (double)tl * fabs((double)wn) +
+          0.5 * (double)tl2 * (double)wn * (double)wn;
   return wn;
 }
 
@@ -682,13 +694,14 @@ template <typename T>
 __global__ __launch_bounds__(BLOCK) void k_prox(int kind, const T* __restrict__ w,
                                                 const T* __restrict__ g,
                                                 double step, double lam,
+                                                double lam2,
                                                 T* __restrict__ out,
                                                 double* __restrict__ reg_part,
                                                 ll n) {
   constexpr int VE = 16 / sizeof(T);
   const ll stride = (ll)gridDim.x * BLOCK;
   const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
-  const T ts = (T)step, tl = (T)lam;
+  const T ts = (T)step, tl = (T)lam, tl2 = (T)lam2;
   double racc = 0.0;
   const ll nv = n / VE;
   for (ll i = gid; i < nv; i += stride) {
@@ -696,11 +709,11 @@ __global__ __launch_bounds__(BLOCK) void k_prox(int kind, const T* __restrict__ 
     loadAcc<T, VE>(w + i * VE, wv);
     loadAcc<T, VE>(g + i * VE, gv);
 #pragma unroll
-    for (int k = 0; k < VE; ++k) ov[k] = prox_elem(kind, wv[k], gv[k], ts, tl, racc);
+    for (int k = 0; k < VE; ++k) ov[k] = prox_elem(kind, wv[k], gv[k], ts, tl, tl2, racc);
     storeAcc<T, VE>(out + i * VE, ov);
   }
   const ll t = nv * VE + gid;
-  if (t < n) out[t] = prox_elem(kind, w[t], g[t], ts, tl, racc);
+  if (t < n) out[t] = prox_elem(kind, w[t], g[t], ts, tl, tl2, racc);
   if (kind != PROX_SIMPLE) {
     double acc[1] = {racc};
     block_reduce_partial<1>(acc, reg_part);
@@ -1112,17 +1125,17 @@ extern "C" int agd_axpby(double a, const void* x, double b, const void* y,
 
 // reg (double[1]) must be zeroed by the caller.
 extern "C" int agd_prox(int kind, const void* w, const void* g, double step,
-                        double lam, void* out, void* reg, long long n,
-                        int dtype, void* red_ws, void* stream) {
+                        double lam, double lam2, void* out, void* reg,
+                        long long n, int dtype, void* red_ws, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const int grid = grid_for(n, BLOCK * 4);
   if (dtype == 1)
     hipLaunchKernelGGL((k_prox<float>), dim3(grid), dim3(BLOCK), 0, s, kind,
-                       (const float*)w, (const float*)g, step, lam, (float*)out,
-                       (double*)red_ws, n);
+                       (const float*)w, (const float*)g, step, lam, lam2,
+                       (float*)out, (double*)red_ws, n);
   else if (dtype == 2)
     hipLaunchKernelGGL((k_prox<double>), dim3(grid), dim3(BLOCK), 0, s, kind,
-                       (const double*)w, (const double*)g, step, lam,
+                       (const double*)w, (const double*)g, step, lam, lam2,
                        (double*)out, (double*)red_ws, n);
   else {
     snprintf(g_err, sizeof(g_err), "agd_prox: bad dtype %d", dtype);

@@ -153,6 +153,16 @@ class CSRShard:
                                          csc=self.csc, need_grad=need_grad)
 
 
+def add_intercept(shard: DenseShard) -> DenseShard:
+    """Return a new DenseShard with an all-ones intercept column prepended
+    (the reference suite's manual pattern, ``Suite.scala:47-49``). Copies the
+    features (n x (d+1))."""
+    ones = torch.ones((shard.n, 1), dtype=shard.features.dtype,
+                      device=shard.features.device)
+    return DenseShard(torch.cat([ones, shard.features], dim=1).contiguous(),
+                      shard.labels)
+
+
 # ---------------------------------------------------------------------------
 # Seeded synthetic generators
 # ---------------------------------------------------------------------------

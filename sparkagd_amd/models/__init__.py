@@ -8,7 +8,7 @@ batched GPU kernels, not per-example JVM loops.
 """
 
 from .gradient import Gradient, LogisticGradient, LeastSquaresGradient, HingeGradient, SmoothedHingeGradient
-from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater
+from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater, ElasticNetUpdater
 from .trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD,
                        regularization_path)
 
@@ -22,6 +22,7 @@ __all__ = [
     "SimpleUpdater",
     "L1Updater",
     "SquaredL2Updater",
+    "ElasticNetUpdater",
     "LogisticRegressionWithAGD",
     "LinearRegressionWithAGD",
     "SVMWithAGD",

@@ -79,6 +79,34 @@ class L1Updater(Updater):
     PROX_KIND = ops.PROX_L1
 
 
+class ElasticNetUpdater(Updater):
+    """Elastic-net prox (beyond the reference's updater family):
+    reg(w) = l1_ratio*lambda*||w||_1 + (1-l1_ratio)*lambda/2*||w||^2;
+    prox = soft-threshold by step*lambda1, then shrink by 1/(1+step*lambda2).
+    Non-affine (margin tracking / Gram solver fall back to the direct path)."""
+
+    PROX_KIND = ops.PROX_ELASTIC_NET
+    AFFINE_PROX = False
+
+    def __init__(self, l1_ratio: float = 0.5):
+        if not (0.0 <= l1_ratio <= 1.0):
+            raise ValueError("l1_ratio must be in [0, 1]")
+        self.l1_ratio = l1_ratio
+
+    def compute(self, weights_old, gradient, step_size, iter, reg_param):  # noqa: A002
+        this_step = step_size / math.sqrt(iter)
+        return ops.prox(self.PROX_KIND, weights_old, gradient, this_step,
+                        self.l1_ratio * reg_param,
+                        (1.0 - self.l1_ratio) * reg_param)
+
+    def reg_value(self, weights, reg_param):
+        zero_grad = torch.zeros_like(weights)
+        _, reg = ops.prox(self.PROX_KIND, weights, zero_grad, 0.0,
+                          self.l1_ratio * reg_param,
+                          (1.0 - self.l1_ratio) * reg_param)
+        return reg
+
+
 class SquaredL2Updater(Updater):
     """w' = w*(1 - step*lambda) - step*g; reg = (lambda/2) ||w'||^2."""
 

@@ -29,6 +29,7 @@ from .reference import (
     PROX_SIMPLE,
     PROX_L1,
     PROX_SQUARED_L2,
+    PROX_ELASTIC_NET,
 )
 
 _hip = None
@@ -143,11 +144,12 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
 
 
 def prox(
-    kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float
+    kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float,
+    lam2: float = 0.0,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(w):
-        return _get_hip().prox(kind, w, g, step, lam)
-    return reference.prox(kind, w, g, step, lam)
+        return _get_hip().prox(kind, w, g, step, lam, lam2)
+    return reference.prox(kind, w, g, step, lam, lam2)
 
 
 def axpby(
@@ -182,6 +184,7 @@ __all__ = [
     "PROX_SIMPLE",
     "PROX_L1",
     "PROX_SQUARED_L2",
+    "PROX_ELASTIC_NET",
     "dense_eval",
     "csr_eval",
     "dense_margins",

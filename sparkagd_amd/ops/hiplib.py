@@ -61,7 +61,7 @@ def load() -> ctypes.CDLL:
     lib.agd_axpby.restype = I
     lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
     lib.agd_prox.restype = I
-    lib.agd_prox.argtypes = [I, P, P, D, D, P, P, LL, I, P, P]
+    lib.agd_prox.argtypes = [I, P, P, D, D, D, P, P, LL, I, P, P]
     lib.agd_fused_scalars.restype = I
     lib.agd_fused_scalars.argtypes = [P, P, P, P, P, LL, I, P, P]
     lib.agd_dot_diff.restype = I
@@ -400,13 +400,14 @@ def axpby(a: float, x: torch.Tensor, b: float, y: torch.Tensor, out: Optional[to
     return out
 
 
-def prox(kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float) -> Tuple[torch.Tensor, torch.Tensor]:
+def prox(kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float,
+         lam2: float = 0.0) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     out = torch.empty_like(w)
     reg = torch.zeros((), dtype=torch.float64, device=w.device)
     rc = lib.agd_prox(kind, _ptr(w.contiguous()), _ptr(g.contiguous()), float(step),
-                      float(lam), _ptr(out), _ptr(reg), w.numel(), _VEC_DTYPE[w.dtype],
-                      _ptr(_red_ws(w.device)), _stream(w))
+                      float(lam), float(lam2), _ptr(out), _ptr(reg), w.numel(),
+                      _VEC_DTYPE[w.dtype], _ptr(_red_ws(w.device)), _stream(w))
     _check(rc)
     return out, reg
 

@@ -35,6 +35,7 @@ LOSS_SMOOTH_HINGE = 3
 PROX_SIMPLE = 0
 PROX_L1 = 1
 PROX_SQUARED_L2 = 2
+PROX_ELASTIC_NET = 3
 
 
 def _multiplier_and_loss(
@@ -225,6 +226,7 @@ def prox(
     g: torch.Tensor,
     step: float,
     lam: float,
+    lam2: float = 0.0,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """One proximal-gradient step + regularization value.
 
@@ -246,6 +248,13 @@ def prox(
     elif kind == PROX_SQUARED_L2:
         w_new = w * (1.0 - step * lam) - step * g
         reg = 0.5 * lam * (w_new.to(torch.float64) ** 2).sum()
+    elif kind == PROX_ELASTIC_NET:
+        # prox of lam*|w|_1 + lam2/2*|w|^2: soft-threshold then shrink
+        w1 = w - step * g
+        wsoft = torch.sign(w1) * torch.clamp(w1.abs() - lam * step, min=0.0)
+        w_new = wsoft / (1.0 + step * lam2)
+        reg = (lam * w_new.abs().to(torch.float64).sum()
+               + 0.5 * lam2 * (w_new.to(torch.float64) ** 2).sum())
     else:
         raise ValueError(f"unknown prox kind {kind}")
     return w_new, reg

@@ -194,7 +194,7 @@ def test_axpby(dtype):
     torch.testing.assert_close(out, 0.3 * x - 1.7 * y)
 
 
-@pytest.mark.parametrize("kind", [ops.PROX_SIMPLE, ops.PROX_L1, ops.PROX_SQUARED_L2])
+@pytest.mark.parametrize("kind", [ops.PROX_SIMPLE, ops.PROX_L1, ops.PROX_SQUARED_L2, ops.PROX_ELASTIC_NET])
 @pytest.mark.parametrize("dtype", [torch.float32, torch.float64])
 def test_prox(kind, dtype):
     from sparkagd_amd.ops import hiplib, reference
@@ -202,8 +202,9 @@ def test_prox(kind, dtype):
     g = torch.Generator(device=DEV).manual_seed(4)
     w = torch.randn(70001, generator=g, device=DEV, dtype=dtype)
     gr = torch.randn(70001, generator=g, device=DEV, dtype=dtype)
-    out_h, reg_h = hiplib.prox(kind, w, gr, 0.37, 0.21)
-    out_r, reg_r = reference.prox(kind, w, gr, 0.37, 0.21)
+    lam2 = 0.11 if kind == ops.PROX_ELASTIC_NET else 0.0
+    out_h, reg_h = hiplib.prox(kind, w, gr, 0.37, 0.21, lam2)
+    out_r, reg_r = reference.prox(kind, w, gr, 0.37, 0.21, lam2)
     torch.testing.assert_close(out_h, out_r, rtol=1e-6, atol=1e-6)
     # reg tolerance: at f32 the elementwise w' differs from torch by fma
     # contraction (1 ulp); the f64 reg sum of 70k such terms walks ~1e-5 abs

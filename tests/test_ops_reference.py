@@ -151,6 +151,38 @@ def test_updaters_mllib_semantics():
     assert abs(float(reg) - 0.5 * lam * (expected**2).sum()) < 1e-10
 
 
+def test_elastic_net_updater():
+    from sparkagd_amd.models.updater import ElasticNetUpdater
+
+    rng = np.random.default_rng(6)
+    d = 13
+    w = rng.normal(size=d)
+    g = rng.normal(size=d)
+    lam, ratio, s0, it = 0.4, 0.3, 0.7, 4
+    s = s0 / math.sqrt(it)
+    l1, l2 = ratio * lam, (1 - ratio) * lam
+    w1 = w - s * g
+    wsoft = np.sign(w1) * np.maximum(np.abs(w1) - l1 * s, 0.0)
+    expected = wsoft / (1.0 + s * l2)
+    w2, reg = ElasticNetUpdater(l1_ratio=ratio).compute(
+        torch.from_numpy(w), torch.from_numpy(g), s0, it, lam)
+    np.testing.assert_allclose(w2.numpy(), expected, rtol=1e-12)
+    exp_reg = l1 * np.abs(expected).sum() + 0.5 * l2 * (expected**2).sum()
+    assert abs(float(reg) - exp_reg) < 1e-10
+    with pytest.raises(ValueError):
+        ElasticNetUpdater(l1_ratio=1.5)
+
+
+def test_add_intercept():
+    from sparkagd_amd.data import add_intercept
+
+    sh = DenseShard(torch.randn(9, 3, dtype=torch.float64), torch.zeros(9))
+    sh2 = add_intercept(sh)
+    assert sh2.d == 4
+    assert torch.all(sh2.features[:, 0] == 1.0)
+    torch.testing.assert_close(sh2.features[:, 1:], sh.features)
+
+
 def test_fused_scalars():
     rng = np.random.default_rng(5)
     d = 33

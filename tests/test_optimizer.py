@@ -148,6 +148,18 @@ def test_smoothed_hinge_beats_plain_hinge_under_agd():
     assert acc > 0.9
 
 
+def test_elastic_net_agd_converges_and_sparsifies():
+    from sparkagd_amd import ElasticNetUpdater
+    from sparkagd_amd.data import generate_dense_problem
+
+    shard, _ = generate_dense_problem(2000, 60, seed=19, dtype=torch.float64)
+    w0 = torch.zeros(60, dtype=torch.float64)
+    w, h = run(shard, LogisticGradient(), ElasticNetUpdater(l1_ratio=0.9),
+               1e-10, 30, 0.05, w0, 1.0, math.inf, 0.5, 0.9, True)
+    assert h[-1] < h[0]
+    assert int((w == 0).sum()) > 0  # the l1 component sparsifies
+
+
 def test_nan_guard():
     """NaN loss -> warn + clean break (AGD.scala:309-312)."""
     feats = torch.tensor([[1e200, 1e200]], dtype=torch.float64)
