@@ -60,6 +60,13 @@ class AGDConfig:
     #: 'gram' precomputes K = A·A^T and iterates in O(n_local·n_global)
     #: (dense shards + affine prox only; see sparkagd_amd/gram.py).
     solver: str = "direct"
+    #: margin-state tracking on the direct solver (optimizer.run docstring);
+    #: True/'auto' enable when eligible, False forces full-evaluation passes
+    #: (bitwise reproducibility vs checkpoints).
+    track_margins: bool | str = "auto"
+    #: recompute tracked margins from the weight vectors every k iterations
+    #: (0 = never; drift is at fp32 rounding level).
+    margin_refresh_every: int = 0
 
     # Numerical guard below which the simple backtracking test switches to the
     # alternate (cancellation-safe) test (reference ``AGD.scala:234-235,272-278``).

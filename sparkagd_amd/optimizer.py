@@ -566,5 +566,7 @@ class AcceleratedGradientDescent(Optimizer):
             checkpoint_every=self.checkpoint_every,
             resume_from=self.resume_from,
             solver=c.solver,
+            track_margins=c.track_margins,
+            margin_refresh_every=c.margin_refresh_every,
         )
         return weights
