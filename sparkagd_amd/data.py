@@ -32,7 +32,8 @@ class DenseShard:
 
     kind = "dense"
 
-    def __init__(self, features: torch.Tensor, labels: torch.Tensor):
+    def __init__(self, features: torch.Tensor, labels: torch.Tensor,
+                 sample_weight: Optional[torch.Tensor] = None):
         if features.ndim != 2:
             raise ValueError("features must be [n, d]")
         if labels.ndim != 1 or labels.shape[0] != features.shape[0]:
@@ -43,6 +44,14 @@ class DenseShard:
         self.labels = labels.to(device=features.device)
         if self.labels.dtype not in (torch.float32, torch.float64):
             self.labels = self.labels.to(torch.float32)
+        # optional per-example weights (weighted GLM; count becomes sum of
+        # weights so all mean-loss/mean-gradient semantics carry over)
+        self.sample_weight = None
+        if sample_weight is not None:
+            if sample_weight.shape[0] != features.shape[0]:
+                raise ValueError("sample_weight must be [n]")
+            self.sample_weight = sample_weight.to(
+                device=features.device, dtype=torch.float32).contiguous()
 
     @property
     def n(self) -> int:
@@ -62,7 +71,8 @@ class DenseShard:
 
     def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None,
              need_grad: bool = True):
-        return ops.dense_eval(self.features, self.labels, w, loss_type, mask, need_grad)
+        return ops.dense_eval(self.features, self.labels, w, loss_type, mask,
+                              need_grad, self.sample_weight)
 
     # --- margin-state tracking support (one data pass instead of two when
     # the caller already holds A @ w; see optimizer.py 'track_margins') ---
@@ -72,7 +82,8 @@ class DenseShard:
     def eval_from_margins(self, margins: torch.Tensor, loss_type: int,
                           mask: Optional[torch.Tensor] = None, need_grad: bool = True):
         return ops.dense_eval_from_margins(self.features, margins, self.labels,
-                                           loss_type, mask, need_grad)
+                                           loss_type, mask, need_grad,
+                                           self.sample_weight)
 
 
 class CSRShard:
@@ -85,7 +96,8 @@ class CSRShard:
     kind = "csr"
 
     def __init__(self, rowptr: torch.Tensor, col: torch.Tensor, val: torch.Tensor,
-                 labels: torch.Tensor, d: int, deterministic: bool = True):
+                 labels: torch.Tensor, d: int, deterministic: bool = True,
+                 sample_weight: Optional[torch.Tensor] = None):
         self.rowptr = rowptr.contiguous()
         self.col = col.contiguous()
         self.val = val.contiguous()
@@ -93,6 +105,10 @@ class CSRShard:
         if self.labels.dtype not in (torch.float32, torch.float64):
             self.labels = self.labels.to(torch.float32)
         self._d = int(d)
+        self.sample_weight = None
+        if sample_weight is not None:
+            self.sample_weight = sample_weight.to(
+                device=val.device, dtype=torch.float32).contiguous()
         # deterministic=True builds a CSC copy of the shard at construction
         # (2x nnz memory) so the A^T·m pass is a gather instead of an fp32
         # atomic scatter: bitwise-reproducible gradients (SURVEY.md §5,
@@ -141,7 +157,8 @@ class CSRShard:
              need_grad: bool = True):
         return ops.csr_eval(self.rowptr, self.col, self.val, self.labels, w,
                             loss_type, mask, self._d, csc=self.csc,
-                            need_grad=need_grad)
+                            need_grad=need_grad,
+                            sample_weight=self.sample_weight)
 
     def margins(self, v: torch.Tensor) -> torch.Tensor:
         return ops.csr_margins(self.rowptr, self.col, self.val, v)
@@ -150,7 +167,8 @@ class CSRShard:
                           mask: Optional[torch.Tensor] = None, need_grad: bool = True):
         return ops.csr_eval_from_margins(self.rowptr, self.col, self.val, margins,
                                          self.labels, loss_type, mask, self._d,
-                                         csc=self.csc, need_grad=need_grad)
+                                         csc=self.csc, need_grad=need_grad,
+                                         sample_weight=self.sample_weight)
 
 
 def add_intercept(shard: DenseShard) -> DenseShard:

@@ -173,8 +173,11 @@ def run_gram(
     n_local = data.n
     feats = data.features
 
-    # count c (full batch, constant)
-    cvec = torch.tensor([float(n_local)], dtype=torch.float64, device=dev)
+    # count c (full batch, constant; sum of weights for weighted shards)
+    if getattr(data, "sample_weight", None) is not None:
+        cvec = data.sample_weight.to(torch.float64).sum().reshape(1).clone()
+    else:
+        cvec = torch.tensor([float(n_local)], dtype=torch.float64, device=dev)
     comm.allreduce_(cvec)
     c = float(cvec[0])
 

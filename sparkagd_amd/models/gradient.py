@@ -59,7 +59,8 @@ class Gradient:
         """(multiplier, loss_count) from precomputed margins (zero data
         passes) — the Gram solver's n-space evaluation."""
         return ops.dense_multiplier_loss(shard.features, margins, shard.labels,
-                                         self.LOSS_TYPE, mask)
+                                         self.LOSS_TYPE, mask,
+                                         getattr(shard, "sample_weight", None))
 
     # --- MLlib per-example API parity (reference Gradient.compute) ---
     def compute(

@@ -80,10 +80,13 @@ def dense_eval(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(features):
-        return _get_hip().dense_eval(features, labels, w, loss_type, mask, need_grad)
-    return reference.dense_eval(features, labels, w, loss_type, mask, need_grad)
+        return _get_hip().dense_eval(features, labels, w, loss_type, mask,
+                                     need_grad, sample_weight)
+    return reference.dense_eval(features, labels, w, loss_type, mask,
+                                need_grad, sample_weight)
 
 
 def csr_eval(
@@ -97,11 +100,14 @@ def csr_eval(
     d: Optional[int] = None,
     csc=None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(val):
-        return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, csc, need_grad)
+        return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask,
+                                   d, csc, need_grad, sample_weight)
     # the torch reference (sparse_csr @ / .t() @) is already deterministic
-    return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d, need_grad)
+    return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d,
+                              need_grad, sample_weight)
 
 
 def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
@@ -110,16 +116,24 @@ def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
     return reference.dense_margins(features, v)
 
 
-def dense_eval_from_margins(features, margins, labels, loss_type, mask=None, need_grad=True):
+def dense_eval_from_margins(features, margins, labels, loss_type, mask=None,
+                            need_grad=True, sample_weight=None):
     if _use_hip(features):
-        return _get_hip().dense_eval_from_margins(features, margins, labels, loss_type, mask, need_grad)
-    return reference.dense_eval_from_margins(features, margins, labels, loss_type, mask, need_grad)
+        return _get_hip().dense_eval_from_margins(features, margins, labels,
+                                                  loss_type, mask, need_grad,
+                                                  sample_weight)
+    return reference.dense_eval_from_margins(features, margins, labels,
+                                             loss_type, mask, need_grad,
+                                             sample_weight)
 
 
-def dense_multiplier_loss(features, margins, labels, loss_type, mask=None):
+def dense_multiplier_loss(features, margins, labels, loss_type, mask=None,
+                          sample_weight=None):
     if _use_hip(features):
-        return _get_hip().dense_multiplier_loss(features, margins, labels, loss_type, mask)
-    return reference.dense_multiplier_loss(features, margins, labels, loss_type, mask)
+        return _get_hip().dense_multiplier_loss(features, margins, labels,
+                                                loss_type, mask, sample_weight)
+    return reference.dense_multiplier_loss(features, margins, labels,
+                                           loss_type, mask, sample_weight)
 
 
 def dense_grad_from_mult(features, mult):
@@ -135,12 +149,15 @@ def csr_margins(rowptr, col, val, v):
 
 
 def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
-                          mask=None, d=None, csc=None, need_grad=True):
+                          mask=None, d=None, csc=None, need_grad=True,
+                          sample_weight=None):
     if _use_hip(val):
         return _get_hip().csr_eval_from_margins(rowptr, col, val, margins, labels,
-                                                loss_type, mask, d, csc, need_grad)
+                                                loss_type, mask, d, csc,
+                                                need_grad, sample_weight)
     return reference.csr_eval_from_margins(rowptr, col, val, margins, labels,
-                                           loss_type, mask, d, csc, need_grad)
+                                           loss_type, mask, d, csc, need_grad,
+                                           sample_weight)
 
 
 def prox(

@@ -55,9 +55,9 @@ def load() -> ctypes.CDLL:
 
     P, LL, I, D = ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_double
     lib.agd_dense_eval.restype = I
-    lib.agd_dense_eval.argtypes = [P, I, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, I, P, P]
+    lib.agd_dense_eval.argtypes = [P, I, P, P, P, P, LL, LL, P, P, P, P, P, LL, I, I, I, I, I, I, P, P]
     lib.agd_csr_eval.restype = I
-    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, I, P, P]
+    lib.agd_csr_eval.argtypes = [P, P, P, P, P, P, P, LL, LL, LL, P, P, P, P, I, P, P, P, I, I, P, P]
     lib.agd_axpby.restype = I
     lib.agd_axpby.argtypes = [D, P, D, P, P, LL, I, P]
     lib.agd_prox.restype = I
@@ -109,6 +109,14 @@ def _prep_mask(mask: Optional[torch.Tensor], device) -> Optional[torch.Tensor]:
     return mask.contiguous().to(device)
 
 
+def _prep_weights(sw: Optional[torch.Tensor], device) -> Optional[torch.Tensor]:
+    if sw is None:
+        return None
+    if sw.dtype != torch.float32:
+        sw = sw.to(torch.float32)
+    return sw.contiguous().to(device)
+
+
 def dense_eval(
     features: torch.Tensor,
     labels: torch.Tensor,
@@ -116,6 +124,7 @@ def dense_eval(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     assert features.is_cuda and features.is_contiguous() and features.ndim == 2
@@ -145,8 +154,9 @@ def dense_eval(
         grad = part = None
         n_rb = 1
 
+    sw = _prep_weights(sample_weight, features.device)
     rc = lib.agd_dense_eval(
-        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(w),
+        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(sw), _ptr(w),
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins), _ptr(mult),
         _ptr(part), n_rb, loss_type, n_slabs, 1 if need_grad else 0,
         margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),  # nt A-stream: +6-9% measured
@@ -169,7 +179,7 @@ def dense_margins(features: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
     n_slabs = int(lib.agd_margin_slabs(n, d, a_dtype, margins_algo))
     margins = torch.empty(n_slabs * n, dtype=acc, device=features.device)
     rc = lib.agd_dense_eval(
-        _ptr(features), a_dtype, None, None, _ptr(v.contiguous()),
+        _ptr(features), a_dtype, None, None, None, _ptr(v.contiguous()),
         n, d, None, None, _ptr(margins), None, None, 1, 0, n_slabs, 0,
         margins_algo, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
         1, _ptr(_red_ws(features.device)), _stream(features),
@@ -185,6 +195,7 @@ def dense_eval_from_margins(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
     """multiplier/loss (+ A^T·m when need_grad) from precomputed margins."""
     lib = load()
@@ -207,8 +218,9 @@ def dense_eval_from_margins(
     else:
         grad = part = None
         n_rb = 1
+    sw = _prep_weights(sample_weight, features.device)
     rc = lib.agd_dense_eval(
-        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), None,
+        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(sw), None,
         n, d, _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
         _ptr(part), n_rb, loss_type, 1, 1 if need_grad else 0,
         1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
@@ -224,6 +236,7 @@ def dense_multiplier_loss(
     labels: torch.Tensor,
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """(mult, loss_count) from precomputed margins — the Gram solver's
     n-space evaluation (no data pass at all)."""
@@ -239,8 +252,9 @@ def dense_multiplier_loss(
     dev = features.device
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
     mult = torch.empty(n, dtype=acc, device=dev)
+    sw = _prep_weights(sample_weight, features.device)
     rc = lib.agd_dense_eval(
-        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), None,
+        _ptr(features), a_dtype, _ptr(labels), _ptr(mask), _ptr(sw), None,
         n, d, None, _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
         None, 1, loss_type, 1, 0,
         1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
@@ -263,7 +277,7 @@ def dense_grad_from_mult(features: torch.Tensor, mult: torch.Tensor) -> torch.Te
     n_rb = int(lib.agd_dense_rowblocks(n, d, a_dtype))
     part = torch.empty(n_rb * d, dtype=acc, device=dev) if n_rb > 1 else grad
     rc = lib.agd_dense_eval(
-        _ptr(features), a_dtype, None, None, None,
+        _ptr(features), a_dtype, None, None, None, None,
         n, d, _ptr(grad), None, None, _ptr(mult.contiguous()),
         _ptr(part), n_rb, 0, 1, 1,
         1, int(os.environ.get("SPARKAGD_NT_LOADS", "1")),
@@ -284,6 +298,7 @@ def csr_eval(
     d: Optional[int] = None,
     csc: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     assert val.is_cuda and val.dtype == torch.float32
@@ -316,8 +331,9 @@ def csr_eval(
     margins = torch.empty(n, dtype=torch.float32, device=dev)
     mult = torch.empty(n, dtype=torch.float32, device=dev)
 
+    sw = _prep_weights(sample_weight, val.device)
     rc = lib.agd_csr_eval(
-        _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask),
+        _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask), _ptr(sw),
         _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
         _ptr(margins), _ptr(mult), loss_type,
         _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 0,
@@ -350,7 +366,7 @@ def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
     margins = torch.empty(n, dtype=torch.float32, device=val.device)
     rc = lib.agd_csr_eval(
         _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
-        None, None, _ptr(v.contiguous()), n, val.numel(), v.numel(),
+        None, None, None, _ptr(v.contiguous()), n, val.numel(), v.numel(),
         None, None, _ptr(margins), None, 0, None, None, None, 0, 1,
         _ptr(_red_ws(val.device)), _stream(val),
     )
@@ -359,7 +375,8 @@ def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
 
 
 def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
-                          mask=None, d=None, csc=None, need_grad=True):
+                          mask=None, d=None, csc=None, need_grad=True,
+                          sample_weight=None):
     lib = load()
     n = rowptr.numel() - 1
     d = d if d is not None else 0
@@ -376,9 +393,10 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
         grad = torch.empty(d, dtype=torch.float32, device=dev); cp, cr, cv = csc
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
     mult = torch.empty(n, dtype=torch.float32, device=dev)
+    sw = _prep_weights(sample_weight, val.device)
     rc = lib.agd_csr_eval(
         _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
-        _ptr(labels), _ptr(mask), None, n, val.numel(), d,
+        _ptr(labels), _ptr(mask), _ptr(sw), None, n, val.numel(), d,
         _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
         loss_type, _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 2,
         _ptr(_red_ws(val.device)), _stream(val),

@@ -71,6 +71,28 @@ def _multiplier_and_loss(
     return mult, loss
 
 
+def _apply_mask_weight(mult, loss, mask, sample_weight, n, device):
+    """Apply the mini-batch mask and/or per-example weights; count is the
+    number of unmasked examples (sum of their weights when weighted)."""
+    if mask is not None:
+        m = mask.to(mult.dtype)
+        mult = mult * m
+        loss = loss * m
+    if sample_weight is not None:
+        sw = sample_weight.to(mult.dtype)
+        mult = mult * sw
+        loss = loss * sw
+        if mask is not None:
+            count = (mask.to(torch.float64) * sample_weight.to(torch.float64)).sum()
+        else:
+            count = sample_weight.to(torch.float64).sum()
+    elif mask is not None:
+        count = mask.sum().to(torch.float64)
+    else:
+        count = torch.tensor(float(n), dtype=torch.float64, device=device)
+    return mult, loss, count
+
+
 def dense_eval(
     features: torch.Tensor,
     labels: torch.Tensor,
@@ -78,6 +100,7 @@ def dense_eval(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Batched loss/gradient over a dense shard.
 
@@ -92,13 +115,8 @@ def dense_eval(
     wa = w.to(acc_dtype)
     z = (features.to(acc_dtype) @ wa) if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else (features @ wa)
     mult, loss = _multiplier_and_loss(z, labels, loss_type)
-    if mask is not None:
-        m = mask.to(mult.dtype)
-        mult = mult * m
-        loss = loss * m
-        count = mask.sum().to(torch.float64)
-    else:
-        count = torch.tensor(float(features.shape[0]), dtype=torch.float64, device=features.device)
+    mult, loss, count = _apply_mask_weight(mult, loss, mask, sample_weight,
+                                           features.shape[0], features.device)
     loss_count = torch.stack([loss.to(torch.float64).sum(), count])
     if not need_grad:
         return None, loss_count
@@ -116,6 +134,7 @@ def csr_eval(
     mask: Optional[torch.Tensor] = None,
     d: Optional[int] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Batched loss/gradient over a CSR shard (MLlib sparse-Vector path analog)."""
     n = rowptr.numel() - 1
@@ -126,13 +145,7 @@ def csr_eval(
     )
     z = a @ w.to(acc_dtype)
     mult, loss = _multiplier_and_loss(z, labels, loss_type)
-    if mask is not None:
-        m = mask.to(mult.dtype)
-        mult = mult * m
-        loss = loss * m
-        count = mask.sum().to(torch.float64)
-    else:
-        count = torch.tensor(float(n), dtype=torch.float64, device=val.device)
+    mult, loss, count = _apply_mask_weight(mult, loss, mask, sample_weight, n, val.device)
     loss_count = torch.stack([loss.to(torch.float64).sum(), count])
     if not need_grad:
         return None, loss_count
@@ -153,16 +166,12 @@ def dense_eval_from_margins(
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
     need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
     acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
     mult, loss = _multiplier_and_loss(margins, labels, loss_type)
-    if mask is not None:
-        m = mask.to(mult.dtype)
-        mult = mult * m
-        loss = loss * m
-        count = mask.sum().to(torch.float64)
-    else:
-        count = torch.tensor(float(features.shape[0]), dtype=torch.float64, device=features.device)
+    mult, loss, count = _apply_mask_weight(mult, loss, mask, sample_weight,
+                                           features.shape[0], features.device)
     loss_count = torch.stack([loss.to(torch.float64).sum(), count])
     if not need_grad:
         return None, loss_count
@@ -176,15 +185,11 @@ def dense_multiplier_loss(
     labels: torch.Tensor,
     loss_type: int,
     mask: Optional[torch.Tensor] = None,
+    sample_weight: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     mult, loss = _multiplier_and_loss(margins, labels, loss_type)
-    if mask is not None:
-        m = mask.to(mult.dtype)
-        mult = mult * m
-        loss = loss * m
-        count = mask.sum().to(torch.float64)
-    else:
-        count = torch.tensor(float(features.shape[0]), dtype=torch.float64, device=features.device)
+    mult, loss, count = _apply_mask_weight(mult, loss, mask, sample_weight,
+                                           features.shape[0], features.device)
     return mult, torch.stack([loss.to(torch.float64).sum(), count])
 
 
@@ -201,16 +206,11 @@ def csr_margins(rowptr, col, val, v: torch.Tensor, n: Optional[int] = None) -> t
 
 
 def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
-                          mask=None, d=None, csc=None, need_grad=True):
+                          mask=None, d=None, csc=None, need_grad=True,
+                          sample_weight=None):
     n = rowptr.numel() - 1
     mult, loss = _multiplier_and_loss(margins, labels, loss_type)
-    if mask is not None:
-        m = mask.to(mult.dtype)
-        mult = mult * m
-        loss = loss * m
-        count = mask.sum().to(torch.float64)
-    else:
-        count = torch.tensor(float(n), dtype=torch.float64, device=val.device)
+    mult, loss, count = _apply_mask_weight(mult, loss, mask, sample_weight, n, val.device)
     loss_count = torch.stack([loss.to(torch.float64).sum(), count])
     if not need_grad:
         return None, loss_count

@@ -73,6 +73,18 @@ def test_dense_eval_f64():
     torch.testing.assert_close(lc_h, lc_r, rtol=1e-9, atol=1e-9)
 
 
+def test_dense_eval_sample_weight():
+    from sparkagd_amd.ops import hiplib, reference
+
+    A, y, w = _mk_dense(5000, 64, torch.float32, seed=6)
+    g = torch.Generator(device=DEV).manual_seed(10)
+    sw = torch.rand(5000, generator=g, device=DEV) * 2.0
+    grad_h, lc_h = hiplib.dense_eval(A, y, w, ops.LOSS_LOGISTIC, sample_weight=sw)
+    grad_r, lc_r = reference.dense_eval(A, y, w, ops.LOSS_LOGISTIC, sample_weight=sw)
+    torch.testing.assert_close(grad_h, grad_r, rtol=1e-4, atol=1e-4)
+    torch.testing.assert_close(lc_h, lc_r, rtol=1e-6, atol=1e-6)
+
+
 def test_dense_eval_masked():
     from sparkagd_amd.ops import hiplib, reference
 
