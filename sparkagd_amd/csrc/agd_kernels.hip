@@ -430,7 +430,8 @@ __device__ __forceinline__ float softplus<float>(float t) {
 template <typename TACC>
 __global__ __launch_bounds__(BLOCK) void k_multiplier(
     const TACC* __restrict__ margins, const float* __restrict__ labels,
-    const unsigned char* __restrict__ mask, int loss_type, ll n, int n_slabs,
+    const unsigned char* __restrict__ mask,
+    const float* __restrict__ sample_weight, int loss_type, ll n, int n_slabs,
     TACC* __restrict__ mult, double* __restrict__ red_part) {
   double lsum = 0.0, cnt = 0.0;
   const ll stride = (ll)gridDim.x * BLOCK;
@@ -469,9 +470,16 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
         l = (TACC)0.5 - sz;
       }
     }
-    mult[i] = m;
-    lsum += (double)l;
-    cnt += 1.0;
+    if (sample_weight) {
+      const TACC sw = (TACC)sample_weight[i];
+      mult[i] = m * sw;
+      lsum += (double)l * (double)sw;
+      cnt += (double)sw;
+    } else {
+      mult[i] = m;
+      lsum += (double)l;
+      cnt += 1.0;
+    }
   }
   double acc[2] = {lsum, cnt};
   block_reduce_partial<2>(acc, red_part);
@@ -868,7 +876,8 @@ extern "C" int agd_margin_slabs(long long n, long long d, int a_dtype,
 
 template <typename TA, typename TACC, int W>
 static int dense_eval_t(const void* A, const float* labels,
-                        const unsigned char* mask, const void* w, ll n, ll d,
+                        const unsigned char* mask, const float* sample_weight,
+                        const void* w, ll n, ll d,
                         void* grad_out, double* loss_count, void* margins_ws,
                         void* mult_ws, void* part_ws, ll n_rb, int loss_type,
                         int n_slabs, int need_grad, int margins_algo,
@@ -925,8 +934,8 @@ static int dense_eval_t(const void* A, const float* labels,
   if (mode != 3) {  // mode 3: the caller provides the multiplier vector
     const int grid = grid_for(n, BLOCK);
     hipLaunchKernelGGL((k_multiplier<TACC>), dim3(grid), dim3(BLOCK), 0, stream,
-                       margins, labels, mask, loss_type, n, n_slabs, mult,
-                       red_ws);
+                       margins, labels, mask, sample_weight, loss_type, n,
+                       n_slabs, mult, red_ws);
     hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, stream,
                        red_ws, grid, loss_count);
   }
@@ -956,7 +965,8 @@ static int dense_eval_t(const void* A, const float* labels,
 // dense shard: margins -> multiplier/loss -> A^T·m partials -> reduce.
 // loss_count[2] must be zeroed by the caller; grad_out is overwritten.
 extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
-                              const void* mask, const void* w, long long n,
+                              const void* mask, const void* sample_weight,
+                              const void* w, long long n,
                               long long d, void* grad_out, void* loss_count,
                               void* margins_ws, void* mult_ws, void* part_ws,
                               long long n_rb, int loss_type, int n_slabs,
@@ -965,17 +975,18 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
   hipStream_t s = (hipStream_t)stream;
   const float* lab = (const float*)labels;
   const unsigned char* msk = (const unsigned char*)mask;
+  const float* swt = (const float*)sample_weight;
   double* lc = (double*)loss_count;
   const int W = pick_w(a_dtype, d);
   switch (a_dtype * 10 + (W > 1 ? 1 : 0)) {
-    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 31: return dense_eval_t<unsigned char, float, 16>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
-    case 30: return dense_eval_t<unsigned char, float, 1>(A, lab, msk, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 1:  return dense_eval_t<ubf16, float, 8>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 0:  return dense_eval_t<ubf16, float, 1>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 11: return dense_eval_t<float, float, 4>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 10: return dense_eval_t<float, float, 1>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 21: return dense_eval_t<double, double, 2>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 20: return dense_eval_t<double, double, 1>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 31: return dense_eval_t<unsigned char, float, 16>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
+    case 30: return dense_eval_t<unsigned char, float, 1>(A, lab, msk, swt, w, n, d, grad_out, lc, margins_ws, mult_ws, part_ws, n_rb, loss_type, n_slabs, need_grad, margins_algo, nt_loads, mode, (double*)red_ws, s);
   }
   snprintf(g_err, sizeof(g_err), "agd_dense_eval: bad dtype %d", a_dtype);
   return 2;
@@ -988,7 +999,8 @@ extern "C" int agd_dense_eval(const void* A, int a_dtype, const void* labels,
 // loss_count must be zeroed by the caller; grad_out additionally so on the
 // atomic path (it is accumulated there, overwritten on the CSC path).
 extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val,
-                            const void* labels, const void* mask, const void* w,
+                            const void* labels, const void* mask,
+                            const void* sample_weight, const void* w,
                             long long n, long long nnz, long long d,
                             void* grad_out, void* loss_count, void* margins_ws,
                             void* mult_ws, int loss_type,
@@ -1015,7 +1027,8 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
     const int grid = grid_for(n, BLOCK);
     hipLaunchKernelGGL((k_multiplier<float>), dim3(grid), dim3(BLOCK), 0, s,
                        margins, (const float*)labels,
-                       (const unsigned char*)mask, loss_type, n, 1, mult,
+                       (const unsigned char*)mask,
+                       (const float*)sample_weight, loss_type, n, 1, mult,
                        (double*)red_ws);
     hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, s,
                        (double*)red_ws, grid, (double*)loss_count);
