@@ -131,10 +131,16 @@ class GramOperator:
     def all_gather_m(self, m_local: torch.Tensor) -> torch.Tensor:
         if self.comm.world_size == 1:
             return m_local
-        parts = [torch.empty(c, dtype=m_local.dtype, device=m_local.device)
-                 for c in self.counts]
-        dist.all_gather(parts, m_local.contiguous())
-        return torch.cat(parts)
+        # NCCL all_gather requires equal lengths: pad to the max count.
+        mx = max(self.counts)
+        buf = m_local
+        if m_local.numel() != mx:
+            buf = torch.zeros(mx, dtype=m_local.dtype, device=m_local.device)
+            buf[: m_local.numel()] = m_local
+        parts = [torch.empty(mx, dtype=m_local.dtype, device=m_local.device)
+                 for _ in self.counts]
+        dist.all_gather(parts, buf.contiguous())
+        return torch.cat([p[:c] for p, c in zip(parts, self.counts)])
 
 
 def run_gram(

@@ -128,6 +128,30 @@ def test_comm_primitives():
         assert all(results[rank]), results[rank]
 
 
+def _dist_gram_uneven(rank, world):
+    # odd global row count -> uneven shards (exercises the padded all_gather)
+    full = generate_logistic_data(2.0, -1.5, N + 1, seed=43)
+    lo, hi = shard_range(N + 1, rank, world)
+    shard = DenseShard(full.features[lo:hi], full.labels[lo:hi])
+    comm = Communicator()
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+    w, hist = run(shard, LogisticGradient(), SquaredL2Updater(), 1e-12, 5, 0.2,
+                  w0, 1.0, math.inf, 0.5, 0.9, True, comm=comm, solver="gram")
+    return w.numpy().tolist(), hist
+
+
+def test_sharded_gram_uneven_rows():
+    results = _run_dist(_dist_gram_uneven, world=2, port=PORT + 5)
+    full = generate_logistic_data(2.0, -1.5, N + 1, seed=43)
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+    w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 5,
+                          0.2, w0, 1.0, math.inf, 0.5, 0.9, True)
+    for rank in (0, 1):
+        torch.testing.assert_close(
+            torch.tensor(results[rank][0], dtype=torch.float64), w_ref,
+            rtol=1e-6, atol=1e-8)
+
+
 def _dist_gram(rank, world):
     shard, _ = _make_shard(rank, world)
     comm = Communicator()
