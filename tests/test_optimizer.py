@@ -174,6 +174,37 @@ def test_nan_guard():
     assert len(hist) < 50  # broke early, did not run to the cap
 
 
+def test_edge_cases():
+    """Degenerate inputs fail safe: zero iterations, single row, d=1."""
+    from sparkagd_amd.data import DenseShard
+
+    data = generate_logistic_data(2.0, -1.5, 100, seed=3)
+    w0 = torch.tensor([0.1, 0.2], dtype=torch.float64)
+
+    # zero iterations: returns the initial weights, empty history
+    w, h = run(data, LogisticGradient(), SimpleUpdater(), 1e-4, 0, 0.0, w0,
+               1.0, math.inf, 0.5, 0.9, True)
+    assert torch.equal(w, w0) and h == []
+
+    # single example
+    one = DenseShard(data.features[:1], data.labels[:1])
+    w, h = run(one, LogisticGradient(), SimpleUpdater(), 1e-10, 5, 0.0, w0,
+               1.0, math.inf, 0.5, 0.9, True)
+    assert len(h) >= 1 and all(map(math.isfinite, h))
+
+    # d = 1
+    narrow = DenseShard(data.features[:, :1].contiguous(), data.labels)
+    w1 = torch.zeros(1, dtype=torch.float64)
+    w, h = run(narrow, LogisticGradient(), SimpleUpdater(), 1e-10, 5, 0.0, w1,
+               1.0, math.inf, 0.5, 0.9, True)
+    assert all(map(math.isfinite, h))
+
+    # mini-batch fraction so small the batch can be empty sometimes: must not crash
+    w, h = run_mini_batch(data, LogisticGradient(), SimpleUpdater(), 1.0, 5,
+                          0.0, 0.01, w0)
+    assert len(h) <= 5
+
+
 def test_restart_fires():
     """The gradient-test restart engages on a poorly conditioned quadratic."""
     import sparkagd_amd.optimizer as om
