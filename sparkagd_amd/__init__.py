@@ -23,6 +23,7 @@ from .models.gradient import (
     LeastSquaresGradient,
     HingeGradient,
     SmoothedHingeGradient,
+    MultinomialLogisticGradient,
 )
 from .models.updater import (
     Updater,
@@ -31,7 +32,8 @@ from .models.updater import (
     SquaredL2Updater,
     ElasticNetUpdater,
 )
-from .data import DenseShard, CSRShard, generate_logistic_data, generate_dense_problem
+from .data import (DenseShard, CSRShard, generate_logistic_data,
+                   generate_dense_problem, generate_multiclass_problem)
 from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniBatch
 from .gram import GramOperator, run_gram
 from .parallel.comm import Communicator
@@ -45,6 +47,7 @@ __all__ = [
     "LeastSquaresGradient",
     "HingeGradient",
     "SmoothedHingeGradient",
+    "MultinomialLogisticGradient",
     "Updater",
     "SimpleUpdater",
     "L1Updater",
@@ -54,6 +57,7 @@ __all__ = [
     "CSRShard",
     "generate_logistic_data",
     "generate_dense_problem",
+    "generate_multiclass_problem",
     "AcceleratedGradientDescent",
     "run",
     "run_mini_batch",

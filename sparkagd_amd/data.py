@@ -261,6 +261,38 @@ def generate_dense_problem(
     return DenseShard(feats, labels), w_true
 
 
+def generate_multiclass_problem(
+    n: int,
+    d: int,
+    num_classes: int,
+    seed: int,
+    device: str | torch.device = "cpu",
+    dtype: torch.dtype = torch.float32,
+    chunk_rows: Optional[int] = None,
+    label_noise: float = 0.5,
+) -> Tuple[DenseShard, torch.Tensor]:
+    """Planted multinomial problem: labels = argmax(X @ W* + noise);
+    returns (shard, w_true_flat [d*K])."""
+    dev = torch.device(device)
+    if chunk_rows is None:
+        chunk_rows = max(64, min(65536, (1 << 28) // max(d, 1)))
+    gen = torch.Generator(device=dev).manual_seed(seed)
+    w_true = torch.randn((d, num_classes), generator=gen, device=dev,
+                         dtype=torch.float32) / math.sqrt(d)
+    feats = torch.empty((n, d), device=dev, dtype=dtype)
+    labels = torch.empty(n, device=dev, dtype=torch.float32)
+    for lo in range(0, n, chunk_rows):
+        hi = min(lo + chunk_rows, n)
+        blk = torch.randn((hi - lo, d), generator=gen, device=dev, dtype=torch.float32)
+        z = blk @ w_true
+        z = z + torch.randn(z.shape, generator=gen, device=dev,
+                            dtype=torch.float32) * label_noise
+        labels[lo:hi] = z.argmax(dim=1).to(torch.float32)
+        feats[lo:hi] = blk.to(dtype)
+        del blk
+    return DenseShard(feats, labels), w_true.reshape(-1)
+
+
 def generate_csr_problem(
     n: int,
     d: int,

@@ -171,6 +171,9 @@ def run_gram(
         raise ValueError("run_gram requires an affine prox updater (Simple/SquaredL2)")
     if getattr(data, "kind", None) != "dense":
         raise ValueError("run_gram requires a DenseShard")
+    if getattr(gradient, "IS_MULTICLASS", False):
+        raise ValueError("run_gram does not support multi-class gradients — "
+                         "use the direct solver")
     backtrack_tol = 1e-10
 
     op = gram_op or GramOperator(data, comm)

@@ -7,7 +7,9 @@ surface is what the reference's tests exercise — but the implementations are
 batched GPU kernels, not per-example JVM loops.
 """
 
-from .gradient import Gradient, LogisticGradient, LeastSquaresGradient, HingeGradient, SmoothedHingeGradient
+from .gradient import (Gradient, LogisticGradient, LeastSquaresGradient,
+                       HingeGradient, SmoothedHingeGradient,
+                       MultinomialLogisticGradient)
 from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater, ElasticNetUpdater
 from .trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD,
                        regularization_path)
@@ -18,6 +20,7 @@ __all__ = [
     "LeastSquaresGradient",
     "HingeGradient",
     "SmoothedHingeGradient",
+    "MultinomialLogisticGradient",
     "Updater",
     "SimpleUpdater",
     "L1Updater",

@@ -113,6 +113,45 @@ class HingeGradient(Gradient):
     LOSS_TYPE = ops.LOSS_HINGE
 
 
+class MultinomialLogisticGradient(Gradient):
+    """Multinomial (softmax) logistic regression over K classes — a model
+    family beyond the reference (MLlib 1.3's LogisticGradient is binary).
+
+    Weights are the flattened [d, K] matrix (feature-major, classes
+    contiguous); labels are class indices 0..K-1; the loss is the softmax
+    cross-entropy; grad = Aᵀ(softmax(Z) − onehot). Dense shards only.
+    Composes with masks, sample weights and margin-state tracking; the Gram
+    solver does not support the multi-class margin structure (use the
+    direct solver)."""
+
+    IS_MULTICLASS = True
+
+    def __init__(self, num_classes: int):
+        if num_classes < 2:
+            raise ValueError("num_classes must be >= 2")
+        self.num_classes = int(num_classes)
+
+    def eval(self, shard, w, mask=None, need_grad=True):
+        from ..ops import multiclass as mc
+
+        if getattr(shard, "kind", None) != "dense":
+            raise ValueError("MultinomialLogisticGradient requires a DenseShard")
+        return mc.eval_multi(shard.features, shard.labels, w, self.num_classes,
+                             mask, need_grad, shard.sample_weight)
+
+    def margins(self, shard, v):
+        from ..ops import multiclass as mc
+
+        return mc.margins_multi(shard.features, v, self.num_classes)
+
+    def eval_from_margins(self, shard, margins, mask=None, need_grad=True):
+        from ..ops import multiclass as mc
+
+        return mc.eval_multi_from_margins(shard.features, margins, shard.labels,
+                                          self.num_classes, mask, need_grad,
+                                          shard.sample_weight)
+
+
 class SmoothedHingeGradient(Gradient):
     """Rennie's quadratically smoothed hinge (differentiable SVM loss):
     loss = 0 if sz>=1; (1-sz)^2/2 if 0<sz<1; 0.5-sz otherwise. Capability

@@ -1,0 +1,130 @@
+"""Multinomial (softmax) logistic regression ops — oracle + dispatch.
+
+A model family beyond the reference (MLlib 1.3's LogisticGradient is binary):
+weights W [d, K] (stored flattened feature-major, classes contiguous),
+margins Z [n, K] = A @ W, labels are class indices in {0..K-1}:
+
+  p_i = softmax(z_i);  loss_i = logsumexp(z_i) - z_{i, y_i}
+  M[i, k] = p_{i,k} - 1[y_i == k];  grad = Aᵀ M   (shape [d, K])
+
+The GPU kernels work on a class dimension padded to KC = ceil(K/4)*4 (16-B
+alignment of the per-feature class row); padding classes carry -inf margins
+conceptually — implemented by restricting every softmax/loss loop to the
+logical K, with padded multiplier columns exactly zero. Padding/unpadding
+happens in this layer; callers only see the logical [d*K] / [n*K] flats.
+
+Composes with the existing machinery: mini-batch masks, per-example sample
+weights, and margin-state tracking (margins are linear in W and the prox
+operators act elementwise on the flattened weights).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import _use_hip, reference as _ref  # noqa: F401  (dispatch helper)
+
+
+def _pad_classes(t2d: torch.Tensor, kc: int) -> torch.Tensor:
+    n, k = t2d.shape
+    if k == kc:
+        return t2d.contiguous()
+    out = torch.zeros((n, kc), dtype=t2d.dtype, device=t2d.device)
+    out[:, :k] = t2d
+    return out
+
+
+def padded_k(k: int) -> int:
+    return (k + 3) // 4 * 4
+
+
+# --- oracle (plain torch; CPU tier + GPU-kernel ground truth) ---
+
+def ref_margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int) -> torch.Tensor:
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
+    w = wflat.reshape(features.shape[1], k).to(acc)
+    return (features.to(acc) @ w).reshape(-1)  # flat [n*k]
+
+
+def ref_eval_multi_from_margins(
+    margins_flat: torch.Tensor,
+    labels: torch.Tensor,
+    features: torch.Tensor,
+    k: int,
+    mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
+) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
+    n = features.shape[0]
+    z = margins_flat.reshape(n, k)
+    acc = z.dtype
+    y = labels.to(torch.int64)
+    lse = torch.logsumexp(z, dim=1)
+    zy = z.gather(1, y.unsqueeze(1)).squeeze(1)
+    loss = lse - zy
+    m = torch.softmax(z, dim=1)
+    m = m.scatter_add(1, y.unsqueeze(1), -torch.ones((n, 1), dtype=acc, device=z.device))
+    scale = torch.ones(n, dtype=acc, device=z.device)
+    count_t = None
+    if mask is not None:
+        scale = scale * mask.to(acc)
+    if sample_weight is not None:
+        scale = scale * sample_weight.to(acc)
+    if mask is not None or sample_weight is not None:
+        loss = loss * scale
+        m = m * scale.unsqueeze(1)
+        count_t = scale.to(torch.float64).sum()
+    if count_t is None:
+        count_t = torch.tensor(float(n), dtype=torch.float64, device=z.device)
+    loss_count = torch.stack([loss.to(torch.float64).sum(), count_t])
+    if not need_grad:
+        return None, loss_count
+    facc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16, torch.float8_e4m3fn) else features.dtype
+    grad = (features.to(facc).T @ m.to(facc)).reshape(-1)  # flat [d*k]
+    return grad, loss_count
+
+
+def ref_eval_multi(features, labels, wflat, k, mask=None, need_grad=True,
+                   sample_weight=None):
+    zf = ref_margins_multi(features, wflat, k)
+    return ref_eval_multi_from_margins(zf, labels, features, k, mask,
+                                       need_grad, sample_weight)
+
+
+# --- dispatch (GPU -> HIP kernels via hiplib; CPU -> oracle) ---
+
+def margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int) -> torch.Tensor:
+    """Flat padded margins [n*KC] (opaque to callers; feed back into
+    eval_multi_from_margins / axpby for margin tracking)."""
+    kc = padded_k(k)
+    if _use_hip(features):
+        from . import hiplib
+
+        return hiplib.dense_margins_multi(features, wflat, k, kc)
+    z = ref_margins_multi(features, wflat, k).reshape(features.shape[0], k)
+    return _pad_classes(z, kc).reshape(-1)
+
+
+def eval_multi_from_margins(features, margins_padded_flat, labels, k,
+                            mask=None, need_grad=True, sample_weight=None):
+    kc = padded_k(k)
+    if _use_hip(features):
+        from . import hiplib
+
+        return hiplib.dense_eval_multi_from_margins(
+            features, margins_padded_flat, labels, k, kc, mask, need_grad,
+            sample_weight)
+    n = features.shape[0]
+    z = margins_padded_flat.reshape(n, kc)[:, :k].reshape(-1)
+    grad, lc = ref_eval_multi_from_margins(z, labels, features, k, mask,
+                                           need_grad, sample_weight)
+    return grad, lc
+
+
+def eval_multi(features, labels, wflat, k, mask=None, need_grad=True,
+               sample_weight=None):
+    zf = margins_multi(features, wflat, k)
+    return eval_multi_from_margins(features, zf, labels, k, mask, need_grad,
+                                   sample_weight)

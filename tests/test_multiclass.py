@@ -1,0 +1,123 @@
+"""Multinomial (softmax) logistic regression — CPU tier."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+from sparkagd_amd import (
+    MultinomialLogisticGradient,
+    SimpleUpdater,
+    SquaredL2Updater,
+    generate_multiclass_problem,
+    run,
+    run_mini_batch,
+)
+from sparkagd_amd.ops import multiclass as mc
+
+
+def _np_softmax_eval(A, y, Wflat, K):
+    n, d = A.shape
+    W = Wflat.reshape(d, K)
+    Z = A @ W
+    Zs = Z - Z.max(axis=1, keepdims=True)
+    P = np.exp(Zs) / np.exp(Zs).sum(axis=1, keepdims=True)
+    lse = np.log(np.exp(Zs).sum(axis=1)) + Z.max(axis=1)
+    loss = (lse - Z[np.arange(n), y.astype(int)]).sum()
+    M = P.copy()
+    M[np.arange(n), y.astype(int)] -= 1.0
+    return (A.T @ M).reshape(-1), loss
+
+
+def test_oracle_matches_numpy():
+    rng = np.random.default_rng(0)
+    n, d, K = 200, 12, 5
+    A = rng.normal(size=(n, d))
+    y = rng.integers(0, K, n).astype(np.float64)
+    W = rng.normal(size=d * K)
+    g_np, l_np = _np_softmax_eval(A, y, W, K)
+    grad, lc = mc.ref_eval_multi(torch.from_numpy(A), torch.from_numpy(y),
+                                 torch.from_numpy(W), K)
+    np.testing.assert_allclose(grad.numpy(), g_np, rtol=1e-9)
+    assert abs(float(lc[0]) - l_np) < 1e-8 * max(1.0, abs(l_np))
+    assert float(lc[1]) == n
+
+
+def test_oracle_finite_difference():
+    """grad matches a central finite difference of the loss."""
+    rng = np.random.default_rng(1)
+    n, d, K = 40, 5, 3
+    A = torch.from_numpy(rng.normal(size=(n, d)))
+    y = torch.from_numpy(rng.integers(0, K, n).astype(np.float64))
+    W = torch.from_numpy(rng.normal(size=d * K))
+    grad, _ = mc.ref_eval_multi(A, y, W, K)
+    eps = 1e-6
+    for idx in [0, 7, d * K - 1]:
+        Wp, Wm = W.clone(), W.clone()
+        Wp[idx] += eps
+        Wm[idx] -= eps
+        _, lp = mc.ref_eval_multi(A, y, Wp, K, need_grad=False)
+        _, lm = mc.ref_eval_multi(A, y, Wm, K, need_grad=False)
+        fd = (float(lp[0]) - float(lm[0])) / (2 * eps)
+        assert abs(fd - float(grad[idx])) < 1e-4 * max(1.0, abs(fd))
+
+
+def test_padded_margins_roundtrip():
+    """K not a multiple of 4 exercises the class padding."""
+    rng = np.random.default_rng(2)
+    n, d, K = 64, 10, 6  # KC = 8
+    A = torch.from_numpy(rng.normal(size=(n, d)))
+    y = torch.from_numpy(rng.integers(0, K, n).astype(np.float64))
+    W = torch.from_numpy(rng.normal(size=d * K))
+    zf = mc.margins_multi(A, W, K)
+    assert zf.numel() == n * mc.padded_k(K)
+    g1, lc1 = mc.eval_multi_from_margins(A, zf, y, K)
+    g2, lc2 = mc.ref_eval_multi(A, y, W, K)
+    torch.testing.assert_close(g1, g2, rtol=1e-9, atol=1e-10)
+    torch.testing.assert_close(lc1, lc2)
+
+
+def test_agd_multiclass_converges():
+    shard, w_true = generate_multiclass_problem(3000, 20, 4, seed=3,
+                                                dtype=torch.float64,
+                                                label_noise=0.15)
+    K = 4
+    grad = MultinomialLogisticGradient(K)
+    w0 = torch.zeros(20 * K, dtype=torch.float64)
+    w, h = run(shard, grad, SquaredL2Updater(), 1e-10, 60, 0.001, w0,
+               1.0, math.inf, 0.5, 0.9, True)
+    assert h[-1] < 0.65 * h[0]  # log(4) start; noise floor bounds the drop
+    # accuracy of the fitted model
+    Z = shard.features @ w.reshape(20, K)
+    acc = float((Z.argmax(dim=1).to(torch.float32) == shard.labels).float().mean())
+    assert acc > 0.8
+    # margin tracking matches the untracked path (trajectory identity)
+    w_u, h_u = run(shard, grad, SquaredL2Updater(), 1e-10, 60, 0.001, w0,
+                   1.0, math.inf, 0.5, 0.9, True, track_margins=False)
+    assert len(h) == len(h_u)
+    for a, b in zip(h, h_u):
+        assert abs(a - b) < 1e-8 * max(1.0, abs(b))
+
+
+def test_minibatch_multiclass():
+    shard, _ = generate_multiclass_problem(2000, 10, 3, seed=5, dtype=torch.float64)
+    grad = MultinomialLogisticGradient(3)
+    w0 = torch.zeros(30, dtype=torch.float64)
+    w, h = run_mini_batch(shard, grad, SimpleUpdater(), 1.0, 15, 0.0, 0.5, w0)
+    assert h[-1] < h[0]
+
+
+def test_multiclass_guards():
+    from sparkagd_amd.data import generate_csr_problem
+
+    with pytest.raises(ValueError):
+        MultinomialLogisticGradient(1)
+    shard, _ = generate_multiclass_problem(100, 5, 3, seed=6, dtype=torch.float64)
+    w0 = torch.zeros(15, dtype=torch.float64)
+    with pytest.raises(ValueError):
+        run(shard, MultinomialLogisticGradient(3), SimpleUpdater(), 1e-6, 3,
+            0.0, w0, 1.0, math.inf, 0.5, 0.9, True, solver="gram")
+    csr, _ = generate_csr_problem(50, 10, 3, seed=7)
+    with pytest.raises(ValueError):
+        MultinomialLogisticGradient(3).eval(csr, w0)
