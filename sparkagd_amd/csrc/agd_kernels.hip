@@ -555,6 +555,145 @@ __global__ __launch_bounds__(BLOCK) void k_grad_reduce(
 }
 
 // ---------------------------------------------------------------------------
+// Multinomial (softmax) logistic regression kernels.
+//
+// Weights W [d, KC] f32 (feature-major, class-padded to KC = ceil(K/4)*4 for
+// 16-B class rows); margins Z and multipliers M are [n, KC] f32 with padded
+// columns exactly zero. KC is a compile-time template (4/8/16/32) so the
+// per-row class accumulators stay in registers (runtime-indexed arrays go to
+// scratch — guide §5.4 rule 20).
+// ---------------------------------------------------------------------------
+
+// Z[r, :] = A[r, :] @ W — one wave per row, lane-strided features; per
+// element the KC class weights are one/two 16-B loads from the L2/L3-cached
+// W; KC wave reductions finish the row.
+template <typename TA, int W, int KC>
+__global__ __launch_bounds__(BLOCK) void k_margins_multi(
+    const TA* __restrict__ A, const float* __restrict__ Wm, ll n, ll d,
+    float* __restrict__ Z) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
+  const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
+  for (ll r = wave_gid; r < n; r += n_waves) {
+    const TA* __restrict__ row = A + r * d;
+    float acc[KC];
+#pragma unroll
+    for (int k = 0; k < KC; ++k) acc[k] = 0.f;
+    ll c = (ll)lane * W;
+    for (; c + W <= d; c += (ll)WAVE * W) {
+      float v[W];
+      loadW<TA, float, W, true>(row + c, v);
+#pragma unroll
+      for (int e = 0; e < W; ++e) {
+        const float* wr = Wm + (c + e) * KC;
+        float wv[KC];
+        loadAcc<float, KC>(wr, wv);
+#pragma unroll
+        for (int k = 0; k < KC; ++k) acc[k] += v[e] * wv[k];
+      }
+    }
+    if (c < d) {
+#pragma unroll
+      for (int e = 0; e < W; ++e)
+        if (c + e < d) {
+          float v[1];
+          loadW<TA, float, 1>(row + c + e, v);
+          const float* wr = Wm + (c + e) * KC;
+#pragma unroll
+          for (int k = 0; k < KC; ++k) acc[k] += v[0] * wr[k];
+        }
+    }
+#pragma unroll
+    for (int k = 0; k < KC; ++k) {
+      acc[k] = wave_reduce_sum(acc[k]);
+      if (lane == 0) Z[r * KC + k] = acc[k];
+    }
+  }
+}
+
+// M[r, :] = softmax(Z[r, :K]) - onehot(y_r) (times mask/weight); per-row
+// loss = logsumexp - z_y. One thread per row; padded classes stay zero.
+template <int KC>
+__global__ __launch_bounds__(BLOCK) void k_multiplier_multi(
+    const float* __restrict__ Z, const float* __restrict__ labels,
+    const unsigned char* __restrict__ mask,
+    const float* __restrict__ sample_weight, ll n, int K,
+    float* __restrict__ M, double* __restrict__ red_part) {
+  double lsum = 0.0, cnt = 0.0;
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
+    float* __restrict__ mrow = M + i * KC;
+    if (mask && !mask[i]) {
+#pragma unroll
+      for (int k = 0; k < KC; ++k) mrow[k] = 0.f;
+      continue;
+    }
+    float z[KC];
+    loadAcc<float, KC>(Z + i * KC, z);
+    const int y = (int)labels[i];
+    float zmax = -3.0e38f;
+#pragma unroll
+    for (int k = 0; k < KC; ++k)
+      if (k < K && z[k] > zmax) zmax = z[k];
+    float esum = 0.f;
+    float p[KC];
+#pragma unroll
+    for (int k = 0; k < KC; ++k) {
+      p[k] = (k < K) ? __expf(z[k] - zmax) : 0.f;
+      esum += p[k];
+    }
+    const float inv = 1.0f / esum;
+    float scale = 1.0f;
+    if (sample_weight) scale = sample_weight[i];
+#pragma unroll
+    for (int k = 0; k < KC; ++k) {
+      float m = p[k] * inv - ((k == y) ? 1.0f : 0.0f);
+      mrow[k] = (k < K) ? m * scale : 0.f;
+    }
+    const float loss = (logf(esum) + zmax) - ((y >= 0 && y < K) ? z[y] : 0.f);
+    lsum += (double)loss * (double)scale;
+    cnt += (double)scale;
+  }
+  double acc[2] = {lsum, cnt};
+  block_reduce_partial<2>(acc, red_part);
+}
+
+// part[rb, c, :] = sum_{r in rb} A[r, c] * M[r, :] — one thread per feature
+// column (256 features per block, coalesced 512 B row segments for bf16),
+// KC accumulators in registers; a reduce over rb finishes (deterministic,
+// reuses k_grad_reduce on length d*KC).
+template <typename TA, int KC>
+__global__ __launch_bounds__(BLOCK) void k_grad_multi(
+    const TA* __restrict__ A, const float* __restrict__ M, ll n, ll d,
+    ll n_rb, float* __restrict__ part) {
+  const ll n_cs = (d + BLOCK - 1) / BLOCK;
+  for (ll b = blockIdx.x; b < n_rb * n_cs; b += gridDim.x) {
+    const ll rb = b / n_cs;
+    const ll cs = b - rb * n_cs;
+    const ll c = cs * BLOCK + threadIdx.x;
+    if (c >= d) continue;
+    const ll r_lo = rb * n / n_rb;
+    const ll r_hi = (rb + 1) * n / n_rb;
+    float acc[KC];
+#pragma unroll
+    for (int k = 0; k < KC; ++k) acc[k] = 0.f;
+    for (ll r = r_lo; r < r_hi; ++r) {
+      float a[1];
+      loadW<TA, float, 1>(A + r * d + c, a);
+      const float* mr = M + r * KC;
+      float mv[KC];
+      loadAcc<float, KC>(mr, mv);
+#pragma unroll
+      for (int k = 0; k < KC; ++k) acc[k] += a[0] * mv[k];
+    }
+    float* dst = part + rb * d * KC + c * KC;
+#pragma unroll
+    for (int k = 0; k < KC; ++k) dst[k] = acc[k];
+  }
+}
+
+// ---------------------------------------------------------------------------
 // K2: CSR margins + transpose gradient (fp32 values)
 //
 // Wave per row; lanes stride the row's nnz; gathered w reads ride L2/LLC
@@ -1116,6 +1255,125 @@ extern "C" int agd_gemm_bf16f32_nt(const void* A, const void* B, void* C,
   hipblasLtMatmulDescDestroy(op);
   HIP_CHECK(hipGetLastError());
   return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Multinomial host entry points. KC in {4, 8, 16, 32} (K <= 32; larger K is
+// rejected — BACKLOG.md). part workspace: n_rb * d * KC floats.
+// ---------------------------------------------------------------------------
+
+extern "C" long long agd_multi_rowblocks(long long n, long long d, int kc) {
+  const ll n_cs = (d + BLOCK - 1) / BLOCK;
+  ll n_rb = 4096 / n_cs;
+  if (n_rb < 1) n_rb = 1;
+  if (n_rb > n) n_rb = n;
+  while (n_rb > 1 && n_rb * d * kc * 4 > (1LL << 30)) n_rb /= 2;
+  return n_rb;
+}
+
+template <typename TA, int W>
+static int margins_multi_t(const void* A, const float* Wm, ll n, ll d, int kc,
+                           float* Z, hipStream_t s) {
+  const int grid = grid_for(n, WAVES_PER_BLOCK);
+#define LAUNCH_MM(KCV)                                                        \
+  hipLaunchKernelGGL((k_margins_multi<TA, W, KCV>), dim3(grid), dim3(BLOCK),  \
+                     0, s, (const TA*)A, Wm, n, d, Z)
+  switch (kc) {
+    case 4: LAUNCH_MM(4); break;
+    case 8: LAUNCH_MM(8); break;
+    case 16: LAUNCH_MM(16); break;
+    case 32: LAUNCH_MM(32); break;
+    default:
+      snprintf(g_err, sizeof(g_err), "margins_multi: bad KC %d", kc);
+      return 2;
+  }
+#undef LAUNCH_MM
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int agd_margins_multi(const void* A, int a_dtype, const void* Wm,
+                                 long long n, long long d, int kc, void* Z,
+                                 void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const float* wp = (const float*)Wm;
+  float* z = (float*)Z;
+  const int W = pick_w(a_dtype, d);
+  if (a_dtype == 0 && W == 8) return margins_multi_t<ubf16, 8>(A, wp, n, d, kc, z, s);
+  if (a_dtype == 0) return margins_multi_t<ubf16, 1>(A, wp, n, d, kc, z, s);
+  if (a_dtype == 1 && W == 4) return margins_multi_t<float, 4>(A, wp, n, d, kc, z, s);
+  if (a_dtype == 1) return margins_multi_t<float, 1>(A, wp, n, d, kc, z, s);
+  if (a_dtype == 3 && W == 16) return margins_multi_t<unsigned char, 16>(A, wp, n, d, kc, z, s);
+  if (a_dtype == 3) return margins_multi_t<unsigned char, 1>(A, wp, n, d, kc, z, s);
+  snprintf(g_err, sizeof(g_err), "margins_multi: dtype %d unsupported", a_dtype);
+  return 2;
+}
+
+extern "C" int agd_multiplier_multi(const void* Z, const void* labels,
+                                    const void* mask, const void* sample_weight,
+                                    long long n, int k, int kc, void* M,
+                                    void* loss_count, void* red_ws,
+                                    void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = grid_for(n, BLOCK);
+#define LAUNCH_MU(KCV)                                                        \
+  hipLaunchKernelGGL((k_multiplier_multi<KCV>), dim3(grid), dim3(BLOCK), 0,   \
+                     s, (const float*)Z, (const float*)labels,                \
+                     (const unsigned char*)mask, (const float*)sample_weight, \
+                     n, k, (float*)M, (double*)red_ws)
+  switch (kc) {
+    case 4: LAUNCH_MU(4); break;
+    case 8: LAUNCH_MU(8); break;
+    case 16: LAUNCH_MU(16); break;
+    case 32: LAUNCH_MU(32); break;
+    default:
+      snprintf(g_err, sizeof(g_err), "multiplier_multi: bad KC %d", kc);
+      return 2;
+  }
+#undef LAUNCH_MU
+  hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, s,
+                     (double*)red_ws, grid, (double*)loss_count);
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+template <typename TA>
+static int grad_multi_t(const void* A, const float* M, ll n, ll d, int kc,
+                        float* part, ll n_rb, float* grad, hipStream_t s) {
+  const ll n_cs = (d + BLOCK - 1) / BLOCK;
+  const int grid = grid_for(n_rb * n_cs, 1);
+#define LAUNCH_GM(KCV)                                                        \
+  hipLaunchKernelGGL((k_grad_multi<TA, KCV>), dim3(grid), dim3(BLOCK), 0, s,  \
+                     (const TA*)A, M, n, d, n_rb, part)
+  switch (kc) {
+    case 4: LAUNCH_GM(4); break;
+    case 8: LAUNCH_GM(8); break;
+    case 16: LAUNCH_GM(16); break;
+    case 32: LAUNCH_GM(32); break;
+    default:
+      snprintf(g_err, sizeof(g_err), "grad_multi: bad KC %d", kc);
+      return 2;
+  }
+#undef LAUNCH_GM
+  if (n_rb > 1) {
+    const int rgrid = grid_for(d * kc, BLOCK);
+    hipLaunchKernelGGL((k_grad_reduce<float>), dim3(rgrid), dim3(BLOCK), 0, s,
+                       part, n_rb, d * kc, grad);
+  }
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int agd_grad_multi(const void* A, int a_dtype, const void* M,
+                              long long n, long long d, int kc, void* part,
+                              long long n_rb, void* grad, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  float* p = (n_rb == 1) ? (float*)grad : (float*)part;
+  if (a_dtype == 0) return grad_multi_t<ubf16>(A, (const float*)M, n, d, kc, p, n_rb, (float*)grad, s);
+  if (a_dtype == 1) return grad_multi_t<float>(A, (const float*)M, n, d, kc, p, n_rb, (float*)grad, s);
+  if (a_dtype == 3) return grad_multi_t<unsigned char>(A, (const float*)M, n, d, kc, p, n_rb, (float*)grad, s);
+  snprintf(g_err, sizeof(g_err), "grad_multi: dtype %d unsupported", a_dtype);
+  return 2;
 }
 
 extern "C" int agd_axpby(double a, const void* x, double b, const void* y,

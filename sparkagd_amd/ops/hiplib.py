@@ -68,6 +68,14 @@ def load() -> ctypes.CDLL:
     lib.agd_dot_diff.argtypes = [P, P, P, P, P, LL, I, P, P]
     lib.agd_gemm_bf16f32_nt.restype = I
     lib.agd_gemm_bf16f32_nt.argtypes = [P, P, P, LL, LL, LL, ctypes.c_float, P]
+    lib.agd_multi_rowblocks.restype = LL
+    lib.agd_multi_rowblocks.argtypes = [LL, LL, I]
+    lib.agd_margins_multi.restype = I
+    lib.agd_margins_multi.argtypes = [P, I, P, LL, LL, I, P, P]
+    lib.agd_multiplier_multi.restype = I
+    lib.agd_multiplier_multi.argtypes = [P, P, P, P, LL, I, I, P, P, P, P]
+    lib.agd_grad_multi.restype = I
+    lib.agd_grad_multi.argtypes = [P, I, P, LL, LL, I, P, LL, P, P]
 
     _lib = lib
     return lib
@@ -340,6 +348,69 @@ def csr_eval(
         _ptr(_red_ws(val.device)), _stream(val),
     )
     _check(rc)
+    return grad, loss_count
+
+
+def dense_margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int,
+                        kc: int) -> torch.Tensor:
+    """Padded flat margins [n*KC] = A @ pad(W [d,K] -> [d,KC])."""
+    lib = load()
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    if features.dtype == torch.float64:
+        raise NotImplementedError("multi-class GPU path supports bf16/f32/f8 shards")
+    dev = features.device
+    w2 = wflat.reshape(d, k).to(torch.float32)
+    if kc != k:
+        wp = torch.zeros((d, kc), dtype=torch.float32, device=dev)
+        wp[:, :k] = w2
+    else:
+        wp = w2.contiguous()
+    Z = torch.empty(n * kc, dtype=torch.float32, device=dev)
+    rc = lib.agd_margins_multi(_ptr(features), a_dtype, _ptr(wp), n, d, kc,
+                               _ptr(Z), _stream(features))
+    _check(rc)
+    return Z
+
+
+def dense_eval_multi_from_margins(
+    features: torch.Tensor,
+    margins_padded_flat: torch.Tensor,
+    labels: torch.Tensor,
+    k: int,
+    kc: int,
+    mask: Optional[torch.Tensor] = None,
+    need_grad: bool = True,
+    sample_weight: Optional[torch.Tensor] = None,
+) -> Tuple[Optional[torch.Tensor], torch.Tensor]:
+    lib = load()
+    n, d = features.shape
+    a_dtype = _DTYPE_CODE[features.dtype]
+    dev = features.device
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, dev)
+    sw = _prep_weights(sample_weight, dev)
+    M = torch.empty(n * kc, dtype=torch.float32, device=dev)
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    rc = lib.agd_multiplier_multi(_ptr(margins_padded_flat.contiguous()),
+                                  _ptr(labels), _ptr(mask), _ptr(sw), n, k, kc,
+                                  _ptr(M), _ptr(loss_count),
+                                  _ptr(_red_ws(dev)), _stream(features))
+    _check(rc)
+    if not need_grad:
+        return None, loss_count
+    n_rb = int(lib.agd_multi_rowblocks(n, d, kc))
+    gradp = torch.empty(d * kc, dtype=torch.float32, device=dev)
+    part = torch.empty(n_rb * d * kc, dtype=torch.float32, device=dev) if n_rb > 1 else gradp
+    rc = lib.agd_grad_multi(_ptr(features), a_dtype, _ptr(M), n, d, kc,
+                            _ptr(part), n_rb, _ptr(gradp), _stream(features))
+    _check(rc)
+    if kc != k:
+        grad = gradp.reshape(d, kc)[:, :k].reshape(-1).contiguous()
+    else:
+        grad = gradp
     return grad, loss_count
 
 

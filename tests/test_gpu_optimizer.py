@@ -95,6 +95,23 @@ def test_gram_solver_matches_direct_gpu():
     assert num / den < 2e-2, (num, den)
 
 
+def test_multiclass_agd_gpu():
+    from sparkagd_amd import MultinomialLogisticGradient
+    from sparkagd_amd.data import generate_multiclass_problem
+
+    K = 10
+    shard, _ = generate_multiclass_problem(40000, 1024, K, seed=21, device=DEV,
+                                           dtype=torch.bfloat16, label_noise=0.2)
+    grad = MultinomialLogisticGradient(K)
+    w0 = torch.zeros(1024 * K, device=DEV, dtype=torch.float32)
+    w, h = run(shard, grad, SquaredL2Updater(), 1e-10, 15, 0.001, w0,
+               1.0, math.inf, 0.5, 0.9, True, loss_history_mode="backtrack")
+    assert h[-1] < 0.6 * h[0]
+    Z = (shard.features.float() @ w.reshape(1024, K))
+    acc = float((Z.argmax(dim=1).to(torch.float32) == shard.labels).float().mean())
+    assert acc > 0.8
+
+
 def test_agd_on_csr_shard_gpu():
     from sparkagd_amd.data import generate_csr_problem
     from sparkagd_amd import ops
