@@ -37,6 +37,13 @@ def _pad_classes(t2d: torch.Tensor, kc: int) -> torch.Tensor:
 
 
 def padded_k(k: int) -> int:
+    """Class-dim padding: snapped to the GPU kernel template sizes
+    (4/8/16/32); K > 32 still works on the CPU oracle tier (plain ceil-4
+    padding) but is rejected by the GPU kernels (BACKLOG: MFMA GEMM margins
+    for large K)."""
+    for kc in (4, 8, 16, 32):
+        if k <= kc:
+            return kc
     return (k + 3) // 4 * 4
 
 
