@@ -37,6 +37,7 @@ from sparkagd_amd import (  # noqa: E402
     LogisticGradient,
     LeastSquaresGradient,
     HingeGradient,
+    MultinomialLogisticGradient,
     SimpleUpdater,
     SquaredL2Updater,
     generate_dense_problem,
@@ -83,6 +84,10 @@ class CountingGradient:
     def LOSS_TYPE(self):
         return self.inner.LOSS_TYPE
 
+    @property
+    def IS_MULTICLASS(self):
+        return getattr(self.inner, "IS_MULTICLASS", False)
+
 
 def sync(device: torch.device) -> None:
     if device.type == "cuda":
@@ -101,6 +106,8 @@ def main() -> int:
     p.add_argument("--reg", type=float, default=0.0)
     p.add_argument("--csr", action="store_true", help="CSR-sparse shard instead of dense")
     p.add_argument("--nnz-per-row", type=int, default=64)
+    p.add_argument("--classes", type=int, default=0,
+                   help=">0: multinomial softmax regression with K classes")
     p.add_argument("--eps", type=float, default=1e-3, help="relative loss-improvement epsilon for iters-to-eps")
     p.add_argument("--solver", type=str, default="direct", choices=["direct", "gram"],
                    help="gram = dual-space solver (K=A.A^T precompute; O(n_local*n_global) iterations)")
@@ -123,7 +130,14 @@ def main() -> int:
     loss_type, grad_cls = LOSSES[args.loss]
 
     t_gen0 = time.perf_counter()
-    if args.csr:
+    if args.classes > 0:
+        from sparkagd_amd.data import generate_multiclass_problem
+
+        shard, _w_true = generate_multiclass_problem(
+            args.rows, args.d, args.classes, seed=1234 + rank * 7,
+            device=device, dtype=dtype if device.type == "cuda" else torch.float64,
+        )
+    elif args.csr:
         shard, _w_true = generate_csr_problem(
             args.rows, args.d, args.nnz_per_row, seed=1234 + rank * 7,
             loss_type=loss_type, device=device,
@@ -136,9 +150,14 @@ def main() -> int:
     sync(device)
     t_gen = time.perf_counter() - t_gen0
 
-    gradient = CountingGradient(grad_cls())
+    if args.classes > 0:
+        gradient = CountingGradient(MultinomialLogisticGradient(args.classes))
+        w0 = torch.zeros(args.d * args.classes, device=device,
+                         dtype=torch.float32 if device.type == "cuda" else torch.float64)
+    else:
+        gradient = CountingGradient(grad_cls())
+        w0 = torch.zeros(args.d, device=device, dtype=wdtype)
     updater = SquaredL2Updater() if args.reg > 0 else SimpleUpdater()
-    w0 = torch.zeros(args.d, device=device, dtype=wdtype)
 
     gram_op = None
     gram_build_seconds = None
@@ -235,7 +254,8 @@ def main() -> int:
             "dtype": args.dtype if device.type == "cuda" else "f32",
             "data": "synthetic",
             "config": {
-                "model": f"{'csr' if args.csr else 'dense'}_{args.loss}_regression",
+                "model": (f"multinomial{args.classes}_regression" if args.classes > 0
+                          else f"{'csr' if args.csr else 'dense'}_{args.loss}_regression"),
                 "d": args.d,
                 "rows_per_gpu": args.rows,
                 "global_rows": global_rows,
