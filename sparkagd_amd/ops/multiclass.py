@@ -102,13 +102,38 @@ def ref_eval_multi(features, labels, wflat, k, mask=None, need_grad=True,
 
 # --- dispatch (GPU -> HIP kernels via hiplib; CPU -> oracle) ---
 
+def _gemm_margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int,
+                        kc: int) -> torch.Tensor:
+    """Margins as a skinny hipBLASLt GEMM: Z[n,KC] = A[n,d]·bf16(W)[d,KC].
+
+    This is the idiomatic route for GEMM-shaped work — the per-element K-fma
+    VALU kernel is load-issue bound at ~12x the A-stream floor for K=16
+    (profiles/r01_multiclass_trace.txt). W is rounded to bf16 (the standard
+    bf16 mixed-precision compute path; A is already bf16, accumulation f32);
+    SPARKAGD_MULTI_MARGINS=valu selects the exact-f32-weights kernel instead.
+    """
+    from . import hiplib
+
+    n, d = features.shape
+    wt = torch.zeros((kc, d), dtype=torch.bfloat16, device=features.device)
+    wt[:k] = wflat.reshape(d, k).T.to(torch.bfloat16)
+    z = torch.empty((n, kc), dtype=torch.float32, device=features.device)
+    hiplib.gemm_bf16f32_nt(features, wt, z)
+    return z.reshape(-1)
+
+
 def margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int) -> torch.Tensor:
     """Flat padded margins [n*KC] (opaque to callers; feed back into
     eval_multi_from_margins / axpby for margin tracking)."""
+    import os
+
     kc = padded_k(k)
     if _use_hip(features):
         from . import hiplib
 
+        algo = os.environ.get("SPARKAGD_MULTI_MARGINS", "auto")
+        if features.dtype == torch.bfloat16 and algo in ("auto", "gemm"):
+            return _gemm_margins_multi(features, wflat, k, kc)
         return hiplib.dense_margins_multi(features, wflat, k, kc)
     z = ref_margins_multi(features, wflat, k).reshape(features.shape[0], k)
     return _pad_classes(z, kc).reshape(-1)

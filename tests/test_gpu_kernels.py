@@ -180,9 +180,12 @@ def test_dense_eval_loss_only():
 
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
 @pytest.mark.parametrize("k", [3, 8])
-def test_multiclass_kernels_match_oracle(dtype, k):
+def test_multiclass_kernels_match_oracle(dtype, k, monkeypatch):
     """margins_multi + multiplier_multi + grad_multi vs the torch oracle
-    (includes non-multiple-of-4 K -> class padding)."""
+    (includes non-multiple-of-4 K -> class padding). For bf16 shards the
+    default margins path is the hipBLASLt GEMM with bf16-rounded W, so the
+    oracle uses the same rounded W; the f32-weights VALU kernel is checked
+    separately under SPARKAGD_MULTI_MARGINS=valu."""
     from sparkagd_amd.ops import multiclass as mc
     from sparkagd_amd.ops import hiplib
 
@@ -191,9 +194,12 @@ def test_multiclass_kernels_match_oracle(dtype, k):
     A = torch.randn((n, d), generator=g, device=DEV).to(dtype).contiguous()
     y = torch.randint(0, k, (n,), generator=g, device=DEV).to(torch.float32)
     W = (torch.randn(d * k, generator=g, device=DEV) / math.sqrt(d)).contiguous()
+    # oracle weights matching the compute path: the GEMM route rounds W to bf16
+    Wr = W.reshape(d, k).to(torch.bfloat16).to(torch.float32).reshape(-1) \
+        if dtype == torch.bfloat16 else W
 
     grad_h, lc_h = mc.eval_multi(A, y, W, k)
-    grad_r, lc_r = mc.ref_eval_multi(A, y, W, k)
+    grad_r, lc_r = mc.ref_eval_multi(A, y, Wr, k)
     torch.testing.assert_close(grad_h, grad_r, rtol=3e-4, atol=3e-3)
     torch.testing.assert_close(lc_h, lc_r, rtol=1e-5, atol=1e-5)
     # loss-only agrees
@@ -203,9 +209,36 @@ def test_multiclass_kernels_match_oracle(dtype, k):
     mask = (torch.rand(n, generator=g, device=DEV) < 0.5).to(torch.uint8)
     sw = torch.rand(n, generator=g, device=DEV) * 2
     gh, lh = mc.eval_multi(A, y, W, k, mask=mask, sample_weight=sw)
-    gr, lr = mc.ref_eval_multi(A, y, W, k, mask=mask, sample_weight=sw)
+    gr, lr = mc.ref_eval_multi(A, y, Wr, k, mask=mask, sample_weight=sw)
     torch.testing.assert_close(gh, gr, rtol=3e-4, atol=3e-3)
     torch.testing.assert_close(lh, lr, rtol=1e-5, atol=1e-5)
+    # the exact-f32-weights VALU kernel is still selectable and tight
+    monkeypatch.setenv("SPARKAGD_MULTI_MARGINS", "valu")
+    gv, lv = mc.eval_multi(A, y, W, k)
+    grad_f32 = mc.ref_eval_multi(A, y, W, k)
+    torch.testing.assert_close(gv, grad_f32[0], rtol=3e-4, atol=3e-3)
+    torch.testing.assert_close(lv, grad_f32[1], rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("k", [10, 16])
+def test_multiclass_gemm_margins_match_valu(k, monkeypatch):
+    """The GEMM margins path equals the VALU kernel run on bf16-rounded
+    weights (same math, different engine), and is run-to-run deterministic."""
+    from sparkagd_amd.ops import multiclass as mc
+
+    g = torch.Generator(device=DEV).manual_seed(52)
+    n, d = 8192, 768
+    A = torch.randn((n, d), generator=g, device=DEV).to(torch.bfloat16).contiguous()
+    W = (torch.randn(d * k, generator=g, device=DEV) / math.sqrt(d)).contiguous()
+    Wr = W.reshape(d, k).to(torch.bfloat16).to(torch.float32).reshape(-1)
+
+    monkeypatch.setenv("SPARKAGD_MULTI_MARGINS", "gemm")
+    zg = mc.margins_multi(A, W, k)
+    zg2 = mc.margins_multi(A, W, k)
+    assert torch.equal(zg, zg2), "hipBLASLt margins GEMM must be deterministic"
+    monkeypatch.setenv("SPARKAGD_MULTI_MARGINS", "valu")
+    zv = mc.margins_multi(A, Wr, k)
+    torch.testing.assert_close(zg, zv, rtol=1e-4, atol=1e-4)
 
 
 def test_csr_csc_deterministic_vs_atomic():
