@@ -1257,6 +1257,56 @@ extern "C" int agd_gemm_bf16f32_nt(const void* A, const void* B, void* C,
   return 0;
 }
 
+// C[d,kc] (f32, row-major) = A[n,d]^T (bf16, row-major) @ M[n,kc] (bf16,
+// row-major) — the multinomial gradient as a skinny TN GEMM. Col-major
+// mapping: C' [kc,d] = M' [kc,n] (opN, M viewed col-major) · A'^T [n,d]
+// (opT on A' = A viewed col-major [d,n]).
+extern "C" int agd_gemm_bf16f32_tn(const void* A, const void* M, void* C,
+                                   long long n, long long d, long long kc,
+                                   void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  if (g_lt_handle == nullptr) {
+    LT_CHECK(hipblasLtCreate(&g_lt_handle));
+    HIP_CHECK(hipMalloc(&g_lt_ws, LT_WS_BYTES));
+  }
+  hipblasLtMatmulDesc_t op = nullptr;
+  hipblasLtMatrixLayout_t la = nullptr, lb = nullptr, lc = nullptr;
+  LT_CHECK(hipblasLtMatmulDescCreate(&op, HIPBLAS_COMPUTE_32F, HIP_R_32F));
+  const hipblasOperation_t opT = HIPBLAS_OP_T, opN = HIPBLAS_OP_N;
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSA,
+                                           &opN, sizeof(opN)));
+  LT_CHECK(hipblasLtMatmulDescSetAttribute(op, HIPBLASLT_MATMUL_DESC_TRANSB,
+                                           &opT, sizeof(opT)));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, kc, n, kc));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, d, n, d));
+  LT_CHECK(hipblasLtMatrixLayoutCreate(&lc, HIP_R_32F, kc, d, kc));
+  const float alpha = 1.0f, beta = 0.0f;
+
+  hipblasLtMatmulPreference_t pref = nullptr;
+  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &LT_WS_BYTES,
+      sizeof(LT_WS_BYTES)));
+  hipblasLtMatmulHeuristicResult_t heur;
+  int found = 0;
+  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(g_lt_handle, op, la, lb, lc, lc,
+                                           pref, 1, &heur, &found));
+  if (found < 1) {
+    snprintf(g_err, sizeof(g_err), "hipblaslt: no algo for tn %lldx%lldx%lld",
+             d, kc, n);
+    return 3;
+  }
+  LT_CHECK(hipblasLtMatmul(g_lt_handle, op, &alpha, M, la, A, lb, &beta, C, lc,
+                           C, lc, &heur.algo, g_lt_ws, LT_WS_BYTES, s));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  hipblasLtMatrixLayoutDestroy(lc);
+  hipblasLtMatrixLayoutDestroy(lb);
+  hipblasLtMatrixLayoutDestroy(la);
+  hipblasLtMatmulDescDestroy(op);
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
 // ---------------------------------------------------------------------------
 // Multinomial host entry points. KC in {4, 8, 16, 32} (K <= 32; larger K is
 // rejected — BACKLOG.md). part workspace: n_rb * d * KC floats.

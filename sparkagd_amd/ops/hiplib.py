@@ -68,6 +68,8 @@ def load() -> ctypes.CDLL:
     lib.agd_dot_diff.argtypes = [P, P, P, P, P, LL, I, P, P]
     lib.agd_gemm_bf16f32_nt.restype = I
     lib.agd_gemm_bf16f32_nt.argtypes = [P, P, P, LL, LL, LL, ctypes.c_float, P]
+    lib.agd_gemm_bf16f32_tn.restype = I
+    lib.agd_gemm_bf16f32_tn.argtypes = [P, P, P, LL, LL, LL, P]
     lib.agd_multi_rowblocks.restype = LL
     lib.agd_multi_rowblocks.argtypes = [LL, LL, I]
     lib.agd_margins_multi.restype = I
@@ -401,12 +403,25 @@ def dense_eval_multi_from_margins(
     _check(rc)
     if not need_grad:
         return None, loss_count
-    n_rb = int(lib.agd_multi_rowblocks(n, d, kc))
-    gradp = torch.empty(d * kc, dtype=torch.float32, device=dev)
-    part = torch.empty(n_rb * d * kc, dtype=torch.float32, device=dev) if n_rb > 1 else gradp
-    rc = lib.agd_grad_multi(_ptr(features), a_dtype, _ptr(M), n, d, kc,
-                            _ptr(part), n_rb, _ptr(gradp), _stream(features))
-    _check(rc)
+    algo = os.environ.get("SPARKAGD_MULTI_GRAD", "auto")
+    if features.dtype == torch.bfloat16 and algo in ("auto", "gemm"):
+        # grad = A^T·M as a skinny hipBLASLt TN GEMM — the per-element KC-fma
+        # VALU kernel is issue-bound ~3.5x the A-stream floor
+        # (profiles/r01_multiclass_trace_gemm.txt). Multipliers round to bf16
+        # (standard mixed precision; A is bf16, accumulation f32);
+        # SPARKAGD_MULTI_GRAD=valu selects the exact-f32-multiplier kernel.
+        gradp = torch.empty(d * kc, dtype=torch.float32, device=dev)
+        rc = lib.agd_gemm_bf16f32_tn(_ptr(features),
+                                     _ptr(M.to(torch.bfloat16)),
+                                     _ptr(gradp), n, d, kc, _stream(features))
+        _check(rc)
+    else:
+        n_rb = int(lib.agd_multi_rowblocks(n, d, kc))
+        gradp = torch.empty(d * kc, dtype=torch.float32, device=dev)
+        part = torch.empty(n_rb * d * kc, dtype=torch.float32, device=dev) if n_rb > 1 else gradp
+        rc = lib.agd_grad_multi(_ptr(features), a_dtype, _ptr(M), n, d, kc,
+                                _ptr(part), n_rb, _ptr(gradp), _stream(features))
+        _check(rc)
     if kc != k:
         grad = gradp.reshape(d, kc)[:, :k].reshape(-1).contiguous()
     else:

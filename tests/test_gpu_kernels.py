@@ -198,9 +198,16 @@ def test_multiclass_kernels_match_oracle(dtype, k, monkeypatch):
     Wr = W.reshape(d, k).to(torch.bfloat16).to(torch.float32).reshape(-1) \
         if dtype == torch.bfloat16 else W
 
+    def check_grad(gh, gr):
+        if dtype == torch.bfloat16:
+            # GEMM grad rounds the multipliers to bf16 -> norm-relative bound
+            assert float(torch.norm(gh - gr)) < 3e-3 * float(torch.norm(gr))
+        else:
+            torch.testing.assert_close(gh, gr, rtol=3e-4, atol=3e-3)
+
     grad_h, lc_h = mc.eval_multi(A, y, W, k)
     grad_r, lc_r = mc.ref_eval_multi(A, y, Wr, k)
-    torch.testing.assert_close(grad_h, grad_r, rtol=3e-4, atol=3e-3)
+    check_grad(grad_h, grad_r)
     torch.testing.assert_close(lc_h, lc_r, rtol=1e-5, atol=1e-5)
     # loss-only agrees
     _, lc2 = mc.eval_multi(A, y, W, k, need_grad=False)
@@ -210,10 +217,11 @@ def test_multiclass_kernels_match_oracle(dtype, k, monkeypatch):
     sw = torch.rand(n, generator=g, device=DEV) * 2
     gh, lh = mc.eval_multi(A, y, W, k, mask=mask, sample_weight=sw)
     gr, lr = mc.ref_eval_multi(A, y, Wr, k, mask=mask, sample_weight=sw)
-    torch.testing.assert_close(gh, gr, rtol=3e-4, atol=3e-3)
+    check_grad(gh, gr)
     torch.testing.assert_close(lh, lr, rtol=1e-5, atol=1e-5)
-    # the exact-f32-weights VALU kernel is still selectable and tight
+    # the exact-f32 VALU kernels are still selectable and tight
     monkeypatch.setenv("SPARKAGD_MULTI_MARGINS", "valu")
+    monkeypatch.setenv("SPARKAGD_MULTI_GRAD", "valu")
     gv, lv = mc.eval_multi(A, y, W, k)
     grad_f32 = mc.ref_eval_multi(A, y, W, k)
     torch.testing.assert_close(gv, grad_f32[0], rtol=3e-4, atol=3e-3)
@@ -239,6 +247,37 @@ def test_multiclass_gemm_margins_match_valu(k, monkeypatch):
     monkeypatch.setenv("SPARKAGD_MULTI_MARGINS", "valu")
     zv = mc.margins_multi(A, Wr, k)
     torch.testing.assert_close(zg, zv, rtol=1e-4, atol=1e-4)
+
+    # TN grad GEMM: deterministic, and equals the VALU kernel fed the same
+    # bf16-rounded multipliers
+    from sparkagd_amd.ops import hiplib
+
+    y = torch.randint(0, k, (n,), generator=g, device=DEV).to(torch.float32)
+    monkeypatch.setenv("SPARKAGD_MULTI_GRAD", "gemm")
+    g1, lc1 = mc.eval_multi_from_margins(A, zg, y, k)
+    g2, lc2 = mc.eval_multi_from_margins(A, zg, y, k)
+    assert torch.equal(g1, g2), "hipBLASLt TN grad GEMM must be deterministic"
+    torch.testing.assert_close(lc1, lc2)
+    monkeypatch.setenv("SPARKAGD_MULTI_GRAD", "valu")
+    kc = mc.padded_k(k)
+    M = torch.empty(n * kc, dtype=torch.float32, device=DEV)
+    lib = hiplib.load()
+    lc = torch.zeros(2, dtype=torch.float64, device=DEV)
+    rc = lib.agd_multiplier_multi(
+        hiplib._ptr(zg.contiguous()), hiplib._ptr(y.contiguous()), None, None,
+        n, k, kc, hiplib._ptr(M), hiplib._ptr(lc),
+        hiplib._ptr(hiplib._red_ws(DEV)), hiplib._stream(A))
+    assert rc == 0
+    Mr = M.to(torch.bfloat16).to(torch.float32)  # what the GEMM path consumed
+    gradp = torch.empty(A.shape[1] * kc, dtype=torch.float32, device=DEV)
+    n_rb = int(lib.agd_multi_rowblocks(n, A.shape[1], kc))
+    part = torch.empty(n_rb * A.shape[1] * kc, dtype=torch.float32, device=DEV) if n_rb > 1 else gradp
+    rc = lib.agd_grad_multi(hiplib._ptr(A), 0, hiplib._ptr(Mr), n, A.shape[1],
+                            kc, hiplib._ptr(part), n_rb, hiplib._ptr(gradp),
+                            hiplib._stream(A))
+    assert rc == 0
+    gv = gradp.reshape(A.shape[1], kc)[:, :k].reshape(-1).contiguous()
+    torch.testing.assert_close(g1, gv, rtol=3e-4, atol=3e-3)
 
 
 def test_csr_csc_deterministic_vs_atomic():
