@@ -410,11 +410,7 @@ def dense_eval_multi_from_margins(
         # (profiles/r01_multiclass_trace_gemm.txt). Multipliers round to bf16
         # (standard mixed precision; A is bf16, accumulation f32);
         # SPARKAGD_MULTI_GRAD=valu selects the exact-f32-multiplier kernel.
-        gradp = torch.empty(d * kc, dtype=torch.float32, device=dev)
-        rc = lib.agd_gemm_bf16f32_tn(_ptr(features),
-                                     _ptr(M.to(torch.bfloat16)),
-                                     _ptr(gradp), n, d, kc, _stream(features))
-        _check(rc)
+        gradp = gemm_bf16f32_tn(features, M.reshape(n, kc)).reshape(-1)
     else:
         n_rb = int(lib.agd_multi_rowblocks(n, d, kc))
         gradp = torch.empty(d * kc, dtype=torch.float32, device=dev)
@@ -444,6 +440,25 @@ def gemm_bf16f32_nt(A: torch.Tensor, B: torch.Tensor, C: torch.Tensor,
                                  float(beta), _stream(A))
     _check(rc)
     return C
+
+
+def gemm_bf16f32_tn(A: torch.Tensor, M: torch.Tensor) -> torch.Tensor:
+    """grad[d,kc] (f32) = A[n,d]^T (bf16) @ M[n,kc] (bf16-rounded) via
+    hipBLASLt — the multinomial gradient GEMM (fp32 accumulation). Accepts M
+    in f32 (rounded here) or bf16; works for any kc."""
+    lib = load()
+    assert A.dtype == torch.bfloat16
+    n, d = A.shape
+    n2, kc = M.shape
+    assert n == n2
+    if M.dtype != torch.bfloat16:
+        M = M.to(torch.bfloat16)
+    M = M.contiguous()
+    grad = torch.empty((d, kc), dtype=torch.float32, device=A.device)
+    rc = lib.agd_gemm_bf16f32_tn(_ptr(A.contiguous()), _ptr(M), _ptr(grad),
+                                 n, d, kc, _stream(A))
+    _check(rc)
+    return grad
 
 
 def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:

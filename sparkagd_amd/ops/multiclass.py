@@ -37,10 +37,10 @@ def _pad_classes(t2d: torch.Tensor, kc: int) -> torch.Tensor:
 
 
 def padded_k(k: int) -> int:
-    """Class-dim padding: snapped to the GPU kernel template sizes
-    (4/8/16/32); K > 32 still works on the CPU oracle tier (plain ceil-4
-    padding) but is rejected by the GPU kernels (BACKLOG: MFMA GEMM margins
-    for large K)."""
+    """Class-dim padding: snapped to the VALU kernel template sizes
+    (4/8/16/32) for small K; plain ceil-4 padding above 32, where the GPU
+    path runs GEMM-shaped (hipBLASLt margins/grad + torch multiplier stage —
+    bf16 or f32 shards)."""
     for kc in (4, 8, 16, 32):
         if k <= kc:
             return kc
@@ -132,11 +132,64 @@ def margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int) -> torch.
         from . import hiplib
 
         algo = os.environ.get("SPARKAGD_MULTI_MARGINS", "auto")
+        if kc > 32:
+            # VALU kernels are KC-templated <= 32; large K runs GEMM-shaped
+            if features.dtype == torch.bfloat16 and algo != "valu":
+                return _gemm_margins_multi(features, wflat, k, kc)
+            if features.dtype == torch.float32:
+                # rocBLAS via torch.mm is the idiomatic f32 library GEMM
+                z = features @ wflat.reshape(features.shape[1], k).to(torch.float32)
+                return _pad_classes(z, kc).reshape(-1)
+            raise NotImplementedError(
+                f"K={k} > 32 margins need a bf16 (hipBLASLt) or f32 (rocBLAS) shard")
         if features.dtype == torch.bfloat16 and algo in ("auto", "gemm"):
             return _gemm_margins_multi(features, wflat, k, kc)
         return hiplib.dense_margins_multi(features, wflat, k, kc)
     z = ref_margins_multi(features, wflat, k).reshape(features.shape[0], k)
     return _pad_classes(z, kc).reshape(-1)
+
+
+def _eval_large_k_from_margins(features, margins_padded_flat, labels, k, kc,
+                               mask, need_grad, sample_weight):
+    """K > 32 multiplier stage in plain torch on the GPU (softmax over n×K is
+    tiny next to the A streams), grad as the hipBLASLt TN GEMM (bf16 shards)
+    or rocBLAS torch.mm (f32 shards)."""
+    import os
+
+    from . import hiplib
+
+    n, d = features.shape
+    z = margins_padded_flat.reshape(n, kc)[:, :k]
+    y = labels.to(torch.int64)
+    lse = torch.logsumexp(z, dim=1)
+    loss = lse - z.gather(1, y.unsqueeze(1)).squeeze(1)
+    m = torch.softmax(z, dim=1)
+    m = m.scatter_add(1, y.unsqueeze(1),
+                      -torch.ones((n, 1), dtype=z.dtype, device=z.device))
+    count_t = None
+    if mask is not None or sample_weight is not None:
+        scale = torch.ones(n, dtype=z.dtype, device=z.device)
+        if mask is not None:
+            scale = scale * mask.to(z.dtype)
+        if sample_weight is not None:
+            scale = scale * sample_weight.to(z.dtype)
+        loss = loss * scale
+        m = m * scale.unsqueeze(1)
+        count_t = scale.to(torch.float64).sum()
+    if count_t is None:
+        count_t = torch.tensor(float(n), dtype=torch.float64, device=z.device)
+    loss_count = torch.stack([loss.to(torch.float64).sum(), count_t])
+    if not need_grad:
+        return None, loss_count
+    if (features.dtype == torch.bfloat16
+            and os.environ.get("SPARKAGD_MULTI_GRAD", "auto") != "valu"):
+        mp = torch.zeros((n, kc), dtype=torch.float32, device=z.device)
+        mp[:, :k] = m
+        return hiplib.gemm_bf16f32_tn(features, mp).reshape(d, kc)[:, :k].reshape(-1).contiguous(), loss_count
+    if features.dtype == torch.float32:
+        return (features.T @ m).reshape(-1), loss_count
+    raise NotImplementedError(
+        f"K={k} > 32 gradient needs a bf16 (hipBLASLt) or f32 (rocBLAS) shard")
 
 
 def eval_multi_from_margins(features, margins_padded_flat, labels, k,
@@ -145,6 +198,10 @@ def eval_multi_from_margins(features, margins_padded_flat, labels, k,
     if _use_hip(features):
         from . import hiplib
 
+        if kc > 32:
+            return _eval_large_k_from_margins(
+                features, margins_padded_flat, labels, k, kc, mask, need_grad,
+                sample_weight)
         return hiplib.dense_eval_multi_from_margins(
             features, margins_padded_flat, labels, k, kc, mask, need_grad,
             sample_weight)

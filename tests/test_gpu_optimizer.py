@@ -107,6 +107,22 @@ def test_multiclass_agd_gpu():
     w, h = run(shard, grad, SquaredL2Updater(), 1e-10, 15, 0.001, w0,
                1.0, math.inf, 0.5, 0.9, True, loss_history_mode="backtrack")
     assert h[-1] < 0.6 * h[0]
+
+
+def test_multiclass_large_k_agd_gpu():
+    """K > 32 end-to-end (GEMM margins/grad + torch multiplier stage,
+    margin tracking on)."""
+    from sparkagd_amd import MultinomialLogisticGradient
+    from sparkagd_amd.data import generate_multiclass_problem
+
+    K = 50
+    shard, _ = generate_multiclass_problem(20000, 512, K, seed=23, device=DEV,
+                                           dtype=torch.bfloat16, label_noise=0.1)
+    grad = MultinomialLogisticGradient(K)
+    w0 = torch.zeros(512 * K, device=DEV, dtype=torch.float32)
+    w, h = run(shard, grad, SquaredL2Updater(), 1e-10, 15, 0.001, w0,
+               1.0, math.inf, 0.5, 0.9, True, loss_history_mode="backtrack")
+    assert h[-1] < 0.6 * h[0]
     Z = (shard.features.float() @ w.reshape(1024, K))
     acc = float((Z.argmax(dim=1).to(torch.float32) == shard.labels).float().mean())
     assert acc > 0.8
