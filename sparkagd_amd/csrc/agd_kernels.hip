@@ -999,7 +999,8 @@ extern "C" int agd_margin_slabs(long long n, long long d, int a_dtype,
     target = 2048;                               // ~8 blocks/CU
     min_slab = 4096;
   } else {
-    n_units = (n + MARGIN_ROWS - 1) / MARGIN_ROWS;  // waves
+    const int r = (a_dtype == 3) ? MARGIN_ROWS_FP8 : MARGIN_ROWS;
+    n_units = (n + r - 1) / r;  // waves
     target = 16384;
     const int w = pick_w(a_dtype, d);
     min_slab = (ll)WAVE * w * 4;
