@@ -269,22 +269,30 @@ __global__ __launch_bounds__(BLOCK) void k_dense_margins(
     TACC acc[R];
 #pragma unroll
     for (int j = 0; j < R; ++j) acc[j] = (TACC)0;
-    ll c = c_lo + (ll)lane * W;
+    // fp8 (1-byte) rows advance only 16 B per lane-chunk; CH=2 keeps 32 B of
+    // each row per lane per iteration so a wave covers 2 KiB/row (matching
+    // the bf16 pattern's DRAM segment size).
+    constexpr int CH = (sizeof(TA) == 1) ? 2 : 1;
+    ll c = c_lo + (ll)lane * (W * CH);
     if (nr == R) {  // full row group (hot path)
-      for (; c + W <= c_hi; c += (ll)WAVE * W) {
-        TACC wv[W];
-        loadAcc<TACC, W>(w + c, wv);
+      for (; c + W * CH <= c_hi; c += (ll)WAVE * W * CH) {
+        TACC wv[CH][W];
+#pragma unroll
+        for (int ch = 0; ch < CH; ++ch) loadAcc<TACC, W>(w + c + ch * W, wv[ch]);
 #pragma unroll
         for (int j = 0; j < R; ++j) {
-          TACC v[W];
-          loadW<TA, TACC, W, NT>(row0 + (ll)j * d + c, v);
 #pragma unroll
-          for (int k = 0; k < W; ++k) acc[j] += v[k] * wv[k];
+          for (int ch = 0; ch < CH; ++ch) {
+            TACC v[W];
+            loadW<TA, TACC, W, NT>(row0 + (ll)j * d + c + ch * W, v);
+#pragma unroll
+            for (int k = 0; k < W; ++k) acc[j] += v[k] * wv[ch][k];
+          }
         }
       }
       if (c < c_hi) {
 #pragma unroll
-        for (int k = 0; k < W; ++k)
+        for (int k = 0; k < W * CH; ++k)
           if (c + k < c_hi) {
             const TACC wk = w[c + k];
 #pragma unroll
@@ -295,8 +303,8 @@ __global__ __launch_bounds__(BLOCK) void k_dense_margins(
             }
           }
       }
-    } else {  // tail row group
-      for (; c < c_hi; c += (ll)WAVE * W) {
+    } else {  // tail row group (at most one per slab; plain W stride)
+      for (c = c_lo + (ll)lane * W; c < c_hi; c += (ll)WAVE * W) {
 #pragma unroll
         for (int k = 0; k < W; ++k)
           if (c + k < c_hi) {
