@@ -269,30 +269,22 @@ __global__ __launch_bounds__(BLOCK) void k_dense_margins(
     TACC acc[R];
 #pragma unroll
     for (int j = 0; j < R; ++j) acc[j] = (TACC)0;
-    // fp8 (1-byte) rows advance only 16 B per lane-chunk; CH=2 keeps 32 B of
-    // each row per lane per iteration so a wave covers 2 KiB/row (matching
-    // the bf16 pattern's DRAM segment size).
-    constexpr int CH = (sizeof(TA) == 1) ? 2 : 1;
-    ll c = c_lo + (ll)lane * (W * CH);
+    ll c = c_lo + (ll)lane * W;
     if (nr == R) {  // full row group (hot path)
-      for (; c + W * CH <= c_hi; c += (ll)WAVE * W * CH) {
-        TACC wv[CH][W];
-#pragma unroll
-        for (int ch = 0; ch < CH; ++ch) loadAcc<TACC, W>(w + c + ch * W, wv[ch]);
+      for (; c + W <= c_hi; c += (ll)WAVE * W) {
+        TACC wv[W];
+        loadAcc<TACC, W>(w + c, wv);
 #pragma unroll
         for (int j = 0; j < R; ++j) {
+          TACC v[W];
+          loadW<TA, TACC, W, NT>(row0 + (ll)j * d + c, v);
 #pragma unroll
-          for (int ch = 0; ch < CH; ++ch) {
-            TACC v[W];
-            loadW<TA, TACC, W, NT>(row0 + (ll)j * d + c + ch * W, v);
-#pragma unroll
-            for (int k = 0; k < W; ++k) acc[j] += v[k] * wv[ch][k];
-          }
+          for (int k = 0; k < W; ++k) acc[j] += v[k] * wv[k];
         }
       }
       if (c < c_hi) {
 #pragma unroll
-        for (int k = 0; k < W * CH; ++k)
+        for (int k = 0; k < W; ++k)
           if (c + k < c_hi) {
             const TACC wk = w[c + k];
 #pragma unroll
