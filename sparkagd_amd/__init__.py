@@ -34,6 +34,9 @@ from .models.updater import (
 )
 from .data import (DenseShard, CSRShard, generate_logistic_data,
                    generate_dense_problem, generate_multiclass_problem)
+from .models.trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD,
+                              SVMWithAGD, SoftmaxRegressionWithAGD, LinearModel,
+                              MultinomialModel, regularization_path)
 from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniBatch
 from .gram import GramOperator, run_gram
 from .parallel.comm import Communicator
@@ -53,6 +56,13 @@ __all__ = [
     "L1Updater",
     "SquaredL2Updater",
     "ElasticNetUpdater",
+    "LogisticRegressionWithAGD",
+    "LinearRegressionWithAGD",
+    "SVMWithAGD",
+    "SoftmaxRegressionWithAGD",
+    "LinearModel",
+    "MultinomialModel",
+    "regularization_path",
     "DenseShard",
     "CSRShard",
     "generate_logistic_data",

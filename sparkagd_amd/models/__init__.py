@@ -12,6 +12,7 @@ from .gradient import (Gradient, LogisticGradient, LeastSquaresGradient,
                        MultinomialLogisticGradient)
 from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater, ElasticNetUpdater
 from .trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD,
+                       SoftmaxRegressionWithAGD, LinearModel, MultinomialModel,
                        regularization_path)
 
 __all__ = [
@@ -29,5 +30,8 @@ __all__ = [
     "LogisticRegressionWithAGD",
     "LinearRegressionWithAGD",
     "SVMWithAGD",
+    "SoftmaxRegressionWithAGD",
+    "LinearModel",
+    "MultinomialModel",
     "regularization_path",
 ]

@@ -127,6 +127,63 @@ def regularization_path(
     return models
 
 
+class MultinomialModel:
+    """A fitted softmax (multinomial logistic) model: weights [d*K]
+    feature-major. predict returns class indices; predict_proba softmax
+    probabilities [n, K]."""
+
+    def __init__(self, weights: torch.Tensor, loss_history, num_classes: int):
+        self.weights = weights
+        self.loss_history = list(loss_history)
+        self.num_classes = int(num_classes)
+
+    def margins(self, features: torch.Tensor) -> torch.Tensor:
+        acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+        w = self.weights.reshape(features.shape[1], self.num_classes).to(acc)
+        return features.to(acc) @ w  # [n, K]
+
+    def predict(self, features: torch.Tensor) -> torch.Tensor:
+        return self.margins(features).argmax(dim=1).to(torch.float32)
+
+    def predict_proba(self, features: torch.Tensor) -> torch.Tensor:
+        return torch.softmax(self.margins(features), dim=1)
+
+
+class SoftmaxRegressionWithAGD:
+    """Multinomial (softmax) logistic regression trainer — a model family
+    beyond the reference (MLlib 1.3's LogisticGradient is binary-only)."""
+
+    @classmethod
+    def train(
+        cls,
+        data,
+        num_classes: int,
+        num_iterations: int = 100,
+        reg_param: float = 0.0,
+        convergence_tol: float = 1e-4,
+        updater: Optional[Updater] = None,
+        initial_weights: Optional[torch.Tensor] = None,
+        comm: Optional[Communicator] = None,
+        config: Optional[AGDConfig] = None,
+    ) -> MultinomialModel:
+        from .gradient import MultinomialLogisticGradient
+
+        cfg = config or AGDConfig()
+        cfg.num_iterations = num_iterations
+        cfg.reg_param = reg_param
+        cfg.convergence_tol = convergence_tol
+        if updater is None:
+            updater = SquaredL2Updater() if reg_param > 0 else SimpleUpdater()
+        opt = AcceleratedGradientDescent(
+            MultinomialLogisticGradient(num_classes), updater, cfg, comm)
+        if initial_weights is None:
+            wdtype = torch.float64 if data.device.type == "cpu" else torch.float32
+            initial_weights = torch.zeros(data.d * num_classes,
+                                          device=data.device, dtype=wdtype)
+        w = opt.optimize(data, initial_weights)
+        return MultinomialModel(w, opt.loss_history, num_classes)
+
+
 class LogisticRegressionWithAGD(_GLMTrainer):
     GRADIENT_CLS = LogisticGradient
     LINK = "logistic"

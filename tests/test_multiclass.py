@@ -108,6 +108,24 @@ def test_minibatch_multiclass():
     assert h[-1] < h[0]
 
 
+def test_softmax_trainer_and_model():
+    from sparkagd_amd import SoftmaxRegressionWithAGD
+
+    shard, _ = generate_multiclass_problem(3000, 20, 4, seed=9,
+                                           dtype=torch.float64,
+                                           label_noise=0.1)
+    model = SoftmaxRegressionWithAGD.train(shard, num_classes=4,
+                                           num_iterations=60, reg_param=0.001)
+    assert model.loss_history[-1] < model.loss_history[0]
+    pred = model.predict(shard.features)
+    acc = float((pred == shard.labels).float().mean())
+    assert acc > 0.8
+    proba = model.predict_proba(shard.features)
+    assert proba.shape == (3000, 4)
+    torch.testing.assert_close(proba.sum(dim=1),
+                               torch.ones(3000, dtype=proba.dtype))
+
+
 def test_multiclass_guards():
     from sparkagd_amd.data import generate_csr_problem
 
