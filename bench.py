@@ -130,7 +130,14 @@ def main() -> int:
     loss_type, grad_cls = LOSSES[args.loss]
 
     t_gen0 = time.perf_counter()
-    if args.classes > 0:
+    if args.classes > 0 and args.csr:
+        from sparkagd_amd.data import generate_multiclass_csr_problem
+
+        shard, _w_true = generate_multiclass_csr_problem(
+            args.rows, args.d, args.nnz_per_row, args.classes,
+            seed=1234 + rank * 7, device=device,
+        )
+    elif args.classes > 0:
         from sparkagd_amd.data import generate_multiclass_problem
 
         shard, _w_true = generate_multiclass_problem(
@@ -254,7 +261,8 @@ def main() -> int:
             "dtype": args.dtype if device.type == "cuda" else "f32",
             "data": "synthetic",
             "config": {
-                "model": (f"multinomial{args.classes}_regression" if args.classes > 0
+                "model": (f"{'csr_' if args.csr else ''}multinomial{args.classes}_regression"
+                          if args.classes > 0
                           else f"{'csr' if args.csr else 'dense'}_{args.loss}_regression"),
                 "d": args.d,
                 "rows_per_gpu": args.rows,

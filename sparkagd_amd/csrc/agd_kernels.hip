@@ -1427,6 +1427,128 @@ extern "C" int agd_grad_multi(const void* A, int a_dtype, const void* M,
   return 2;
 }
 
+// ---------------------------------------------------------------------------
+// Multinomial on CSR shards. Z[n,KC] = A·W and grad[d,KC] = Aᵀ·M where the
+// per-nonzero work is KC fmas on a 16-B-aligned KC-float row of W (margins)
+// or M (gradient) — the gathered lines carry KC useful floats, so the
+// per-line efficiency is KC/1 versus the binary kernels' single float.
+// One THREAD per row/column (no cross-lane reduction; acc[KC] in registers);
+// the CSC gather keeps the gradient atomics-free => bitwise deterministic.
+// ---------------------------------------------------------------------------
+
+template <int KC>
+__global__ __launch_bounds__(BLOCK) void k_csr_margins_multi(
+    const int* __restrict__ rowptr, const int* __restrict__ col,
+    const float* __restrict__ val, const float* __restrict__ w, ll n,
+    float* __restrict__ Z) {
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll r = (ll)blockIdx.x * BLOCK + threadIdx.x; r < n; r += stride) {
+    const int k_lo = rowptr[r], k_hi = rowptr[r + 1];
+    float acc[KC];
+#pragma unroll
+    for (int j = 0; j < KC; ++j) acc[j] = 0.f;
+    for (int k = k_lo; k < k_hi; ++k) {
+      const float v = val[k];
+      const float* __restrict__ wr = w + (ll)col[k] * KC;
+#pragma unroll
+      for (int ch = 0; ch < KC / 4; ++ch) {
+        const f32x4 wv = *(const f32x4*)(wr + ch * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[ch * 4 + j] += v * wv[j];
+      }
+    }
+    float* __restrict__ zr = Z + r * KC;
+#pragma unroll
+    for (int ch = 0; ch < KC / 4; ++ch) {
+      f32x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = acc[ch * 4 + j];
+      *(f32x4*)(zr + ch * 4) = o;
+    }
+  }
+}
+
+template <int KC>
+__global__ __launch_bounds__(BLOCK) void k_csc_grad_multi(
+    const int* __restrict__ colptr, const int* __restrict__ row,
+    const float* __restrict__ val, const float* __restrict__ M, ll d,
+    float* __restrict__ grad) {
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll c = (ll)blockIdx.x * BLOCK + threadIdx.x; c < d; c += stride) {
+    const int k_lo = colptr[c], k_hi = colptr[c + 1];
+    float acc[KC];
+#pragma unroll
+    for (int j = 0; j < KC; ++j) acc[j] = 0.f;
+    for (int k = k_lo; k < k_hi; ++k) {
+      const float v = val[k];
+      const float* __restrict__ mr = M + (ll)row[k] * KC;
+#pragma unroll
+      for (int ch = 0; ch < KC / 4; ++ch) {
+        const f32x4 mv = *(const f32x4*)(mr + ch * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[ch * 4 + j] += v * mv[j];
+      }
+    }
+    float* __restrict__ gr = grad + c * KC;
+#pragma unroll
+    for (int ch = 0; ch < KC / 4; ++ch) {
+      f32x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = acc[ch * 4 + j];
+      *(f32x4*)(gr + ch * 4) = o;
+    }
+  }
+}
+
+extern "C" int agd_csr_margins_multi(const void* rowptr, const void* col,
+                                     const void* val, const void* w,
+                                     long long n, int kc, void* Z,
+                                     void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = grid_for(n, BLOCK);
+#define LAUNCH_CM(KCV)                                                        \
+  hipLaunchKernelGGL((k_csr_margins_multi<KCV>), dim3(grid), dim3(BLOCK), 0,  \
+                     s, (const int*)rowptr, (const int*)col,                  \
+                     (const float*)val, (const float*)w, n, (float*)Z)
+  switch (kc) {
+    case 4: LAUNCH_CM(4); break;
+    case 8: LAUNCH_CM(8); break;
+    case 16: LAUNCH_CM(16); break;
+    case 32: LAUNCH_CM(32); break;
+    default:
+      snprintf(g_err, sizeof(g_err), "csr_margins_multi: bad KC %d", kc);
+      return 2;
+  }
+#undef LAUNCH_CM
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int agd_csc_grad_multi(const void* colptr, const void* row,
+                                  const void* val, const void* M, long long d,
+                                  int kc, void* grad, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = grid_for(d, BLOCK);
+#define LAUNCH_CG(KCV)                                                        \
+  hipLaunchKernelGGL((k_csc_grad_multi<KCV>), dim3(grid), dim3(BLOCK), 0, s,  \
+                     (const int*)colptr, (const int*)row, (const float*)val,  \
+                     (const float*)M, d, (float*)grad)
+  switch (kc) {
+    case 4: LAUNCH_CG(4); break;
+    case 8: LAUNCH_CG(8); break;
+    case 16: LAUNCH_CG(16); break;
+    case 32: LAUNCH_CG(32); break;
+    default:
+      snprintf(g_err, sizeof(g_err), "csc_grad_multi: bad KC %d", kc);
+      return 2;
+  }
+#undef LAUNCH_CG
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
 extern "C" int agd_axpby(double a, const void* x, double b, const void* y,
                          void* out, long long n, int dtype, void* stream) {
   hipStream_t s = (hipStream_t)stream;

@@ -293,6 +293,39 @@ def generate_multiclass_problem(
     return DenseShard(feats, labels), w_true.reshape(-1)
 
 
+def generate_multiclass_csr_problem(
+    n: int,
+    d: int,
+    nnz_per_row: int,
+    num_classes: int,
+    seed: int,
+    device: str | torch.device = "cpu",
+    label_noise: float = 0.5,
+) -> Tuple[CSRShard, torch.Tensor]:
+    """Planted multinomial CSR problem: labels = argmax(A @ W* + noise);
+    returns (shard, w_true_flat [d*K])."""
+    dev = torch.device(device)
+    gen = torch.Generator(device=dev).manual_seed(seed)
+    nnz = n * nnz_per_row
+    col = torch.randint(0, d, (nnz,), generator=gen, device=dev, dtype=torch.int32)
+    col = col.view(n, nnz_per_row).sort(dim=1).values.reshape(-1).contiguous()
+    val = torch.randn(nnz, generator=gen, device=dev, dtype=torch.float32)
+    rowptr = torch.arange(0, nnz + 1, nnz_per_row, device=dev, dtype=torch.int32)
+    w_true = torch.randn((d, num_classes), generator=gen, device=dev,
+                         dtype=torch.float32) / math.sqrt(nnz_per_row)
+    labels = torch.empty(n, device=dev, dtype=torch.float32)
+    chunk = max(1, (1 << 22) // max(nnz_per_row * num_classes, 1))
+    for lo in range(0, n, chunk):
+        hi = min(lo + chunk, n)
+        c = col[lo * nnz_per_row: hi * nnz_per_row].view(hi - lo, nnz_per_row).to(torch.int64)
+        v = val[lo * nnz_per_row: hi * nnz_per_row].view(hi - lo, nnz_per_row)
+        z = (v.unsqueeze(2) * w_true[c]).sum(dim=1)  # [rows, K]
+        z = z + torch.randn(z.shape, generator=gen, device=dev,
+                            dtype=torch.float32) * label_noise
+        labels[lo:hi] = z.argmax(dim=1).to(torch.float32)
+    return CSRShard(rowptr, col, val, labels, d), w_true.reshape(-1)
+
+
 def generate_csr_problem(
     n: int,
     d: int,

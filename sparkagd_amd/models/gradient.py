@@ -119,7 +119,8 @@ class MultinomialLogisticGradient(Gradient):
 
     Weights are the flattened [d, K] matrix (feature-major, classes
     contiguous); labels are class indices 0..K-1; the loss is the softmax
-    cross-entropy; grad = Aᵀ(softmax(Z) − onehot). Dense shards only.
+    cross-entropy; grad = Aᵀ(softmax(Z) − onehot). Works on dense and CSR
+    shards (CSR gradients use the deterministic CSC gather).
     Composes with masks, sample weights and margin-state tracking; the Gram
     solver does not support the multi-class margin structure (use the
     direct solver)."""
@@ -134,19 +135,27 @@ class MultinomialLogisticGradient(Gradient):
     def eval(self, shard, w, mask=None, need_grad=True):
         from ..ops import multiclass as mc
 
-        if getattr(shard, "kind", None) != "dense":
-            raise ValueError("MultinomialLogisticGradient requires a DenseShard")
+        if getattr(shard, "kind", None) == "csr":
+            return mc.eval_multi_csr(shard, w, self.num_classes, mask,
+                                     need_grad, shard.sample_weight)
         return mc.eval_multi(shard.features, shard.labels, w, self.num_classes,
                              mask, need_grad, shard.sample_weight)
 
     def margins(self, shard, v):
         from ..ops import multiclass as mc
 
+        if getattr(shard, "kind", None) == "csr":
+            return mc.csr_margins_multi(shard, v, self.num_classes)
         return mc.margins_multi(shard.features, v, self.num_classes)
 
     def eval_from_margins(self, shard, margins, mask=None, need_grad=True):
         from ..ops import multiclass as mc
 
+        if getattr(shard, "kind", None) == "csr":
+            return mc.eval_multi_csr_from_margins(shard, margins,
+                                                  self.num_classes, mask,
+                                                  need_grad,
+                                                  shard.sample_weight)
         return mc.eval_multi_from_margins(shard.features, margins, shard.labels,
                                           self.num_classes, mask, need_grad,
                                           shard.sample_weight)

@@ -78,6 +78,10 @@ def load() -> ctypes.CDLL:
     lib.agd_multiplier_multi.argtypes = [P, P, P, P, LL, I, I, P, P, P, P]
     lib.agd_grad_multi.restype = I
     lib.agd_grad_multi.argtypes = [P, I, P, LL, LL, I, P, LL, P, P]
+    lib.agd_csr_margins_multi.restype = I
+    lib.agd_csr_margins_multi.argtypes = [P, P, P, P, LL, I, P, P]
+    lib.agd_csc_grad_multi.restype = I
+    lib.agd_csc_grad_multi.argtypes = [P, P, P, P, LL, I, P, P]
 
     _lib = lib
     return lib
@@ -375,6 +379,34 @@ def dense_margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int,
     return Z
 
 
+def multiplier_multi(
+    margins_padded_flat: torch.Tensor,
+    labels: torch.Tensor,
+    k: int,
+    kc: int,
+    mask: Optional[torch.Tensor] = None,
+    sample_weight: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(M [n*kc] f32, loss_count f64[2]) from padded softmax margins —
+    the n-space multiplier stage shared by the dense and CSR paths."""
+    lib = load()
+    dev = margins_padded_flat.device
+    n = margins_padded_flat.numel() // kc
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mask = _prep_mask(mask, dev)
+    sw = _prep_weights(sample_weight, dev)
+    M = torch.empty(n * kc, dtype=torch.float32, device=dev)
+    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
+    rc = lib.agd_multiplier_multi(_ptr(margins_padded_flat.contiguous()),
+                                  _ptr(labels), _ptr(mask), _ptr(sw), n, k, kc,
+                                  _ptr(M), _ptr(loss_count),
+                                  _ptr(_red_ws(dev)), _stream(margins_padded_flat))
+    _check(rc)
+    return M, loss_count
+
+
 def dense_eval_multi_from_margins(
     features: torch.Tensor,
     margins_padded_flat: torch.Tensor,
@@ -389,18 +421,8 @@ def dense_eval_multi_from_margins(
     n, d = features.shape
     a_dtype = _DTYPE_CODE[features.dtype]
     dev = features.device
-    labels = labels.contiguous()
-    if labels.dtype != torch.float32:
-        labels = labels.to(torch.float32)
-    mask = _prep_mask(mask, dev)
-    sw = _prep_weights(sample_weight, dev)
-    M = torch.empty(n * kc, dtype=torch.float32, device=dev)
-    loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
-    rc = lib.agd_multiplier_multi(_ptr(margins_padded_flat.contiguous()),
-                                  _ptr(labels), _ptr(mask), _ptr(sw), n, k, kc,
-                                  _ptr(M), _ptr(loss_count),
-                                  _ptr(_red_ws(dev)), _stream(features))
-    _check(rc)
+    M, loss_count = multiplier_multi(margins_padded_flat, labels, k, kc, mask,
+                                     sample_weight)
     if not need_grad:
         return None, loss_count
     algo = os.environ.get("SPARKAGD_MULTI_GRAD", "auto")
@@ -423,6 +445,41 @@ def dense_eval_multi_from_margins(
     else:
         grad = gradp
     return grad, loss_count
+
+
+def csr_margins_multi(rowptr, col, val, wflat: torch.Tensor, k: int,
+                      kc: int, d: int) -> torch.Tensor:
+    """Padded flat margins [n*KC] = CSR(A) @ pad(W [d,K] -> [d,KC])."""
+    lib = load()
+    n = rowptr.numel() - 1
+    dev = val.device
+    w2 = wflat.reshape(d, k).to(torch.float32)
+    if kc != k:
+        wp = torch.zeros((d, kc), dtype=torch.float32, device=dev)
+        wp[:, :k] = w2
+    else:
+        wp = w2.contiguous()
+    Z = torch.empty(n * kc, dtype=torch.float32, device=dev)
+    rc = lib.agd_csr_margins_multi(_ptr(rowptr.contiguous()),
+                                   _ptr(col.contiguous()),
+                                   _ptr(val.contiguous()), _ptr(wp), n, kc,
+                                   _ptr(Z), _stream(val))
+    _check(rc)
+    return Z
+
+
+def csc_grad_multi(colptr, row, cval, M: torch.Tensor, d: int,
+                   kc: int) -> torch.Tensor:
+    """grad flat [d*KC] = A^T·M via the deterministic CSC gather."""
+    lib = load()
+    grad = torch.empty(d * kc, dtype=torch.float32, device=cval.device)
+    rc = lib.agd_csc_grad_multi(_ptr(colptr.contiguous()),
+                                _ptr(row.contiguous()),
+                                _ptr(cval.contiguous()),
+                                _ptr(M.contiguous()), d, kc, _ptr(grad),
+                                _stream(cval))
+    _check(rc)
+    return grad
 
 
 def gemm_bf16f32_nt(A: torch.Tensor, B: torch.Tensor, C: torch.Tensor,

@@ -100,6 +100,63 @@ def ref_eval_multi(features, labels, wflat, k, mask=None, need_grad=True,
                                        need_grad, sample_weight)
 
 
+def ref_multiplier_multi(
+    z: torch.Tensor,  # [n, k] logical margins
+    labels: torch.Tensor,
+    mask: Optional[torch.Tensor] = None,
+    sample_weight: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(M [n,k], loss_count f64[2]) — the n-space softmax multiplier stage."""
+    n, k = z.shape
+    acc = z.dtype
+    y = labels.to(torch.int64)
+    lse = torch.logsumexp(z, dim=1)
+    loss = lse - z.gather(1, y.unsqueeze(1)).squeeze(1)
+    m = torch.softmax(z, dim=1)
+    m = m.scatter_add(1, y.unsqueeze(1),
+                      -torch.ones((n, 1), dtype=acc, device=z.device))
+    count_t = None
+    if mask is not None or sample_weight is not None:
+        scale = torch.ones(n, dtype=acc, device=z.device)
+        if mask is not None:
+            scale = scale * mask.to(acc)
+        if sample_weight is not None:
+            scale = scale * sample_weight.to(acc)
+        loss = loss * scale
+        m = m * scale.unsqueeze(1)
+        count_t = scale.to(torch.float64).sum()
+    if count_t is None:
+        count_t = torch.tensor(float(n), dtype=torch.float64, device=z.device)
+    return m, torch.stack([loss.to(torch.float64).sum(), count_t])
+
+
+# --- CSR shards (oracle; GPU kernels are k_csr_margins_multi/k_csc_grad_multi) ---
+
+def _csr_rows(rowptr: torch.Tensor) -> torch.Tensor:
+    n = rowptr.numel() - 1
+    counts = torch.diff(rowptr.to(torch.int64))
+    return torch.repeat_interleave(
+        torch.arange(n, device=rowptr.device, dtype=torch.int64), counts)
+
+
+def ref_csr_margins_multi(rowptr, col, val, wflat, k: int, d: int) -> torch.Tensor:
+    acc = torch.float32 if val.dtype in (torch.bfloat16, torch.float16) else val.dtype
+    w = wflat.reshape(d, k).to(acc)
+    n = rowptr.numel() - 1
+    z = torch.zeros((n, k), dtype=acc, device=val.device)
+    z.index_add_(0, _csr_rows(rowptr),
+                 val.to(acc).unsqueeze(1) * w[col.to(torch.int64)])
+    return z.reshape(-1)
+
+
+def ref_csr_grad_multi(rowptr, col, val, m2d: torch.Tensor, d: int) -> torch.Tensor:
+    acc = m2d.dtype
+    grad = torch.zeros((d, m2d.shape[1]), dtype=acc, device=val.device)
+    grad.index_add_(0, col.to(torch.int64),
+                    val.to(acc).unsqueeze(1) * m2d[_csr_rows(rowptr)])
+    return grad.reshape(-1)
+
+
 # --- dispatch (GPU -> HIP kernels via hiplib; CPU -> oracle) ---
 
 def _gemm_margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int,
@@ -217,3 +274,55 @@ def eval_multi(features, labels, wflat, k, mask=None, need_grad=True,
     zf = margins_multi(features, wflat, k)
     return eval_multi_from_margins(features, zf, labels, k, mask, need_grad,
                                    sample_weight)
+
+
+# --- CSR dispatch ---
+
+def csr_margins_multi(shard, wflat: torch.Tensor, k: int) -> torch.Tensor:
+    """Flat padded margins [n*KC] over a CSRShard."""
+    kc = padded_k(k)
+    if _use_hip(shard.val):
+        from . import hiplib
+
+        if kc > 32:
+            raise NotImplementedError(
+                f"CSR multiclass GPU margins support K <= 32 (K={k})")
+        return hiplib.csr_margins_multi(shard.rowptr, shard.col, shard.val,
+                                        wflat, k, kc, shard.d)
+    z = ref_csr_margins_multi(shard.rowptr, shard.col, shard.val, wflat, k,
+                              shard.d).reshape(-1, k)
+    return _pad_classes(z, kc).reshape(-1)
+
+
+def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
+                                mask=None, need_grad=True, sample_weight=None):
+    kc = padded_k(k)
+    n = shard.n
+    if _use_hip(shard.val):
+        from . import hiplib
+
+        M, lc = hiplib.multiplier_multi(margins_padded_flat, shard.labels, k,
+                                        kc, mask, sample_weight)
+        if not need_grad:
+            return None, lc
+        if shard.csc is None:
+            raise RuntimeError(
+                "CSR multiclass gradient needs the deterministic CSC copy "
+                "(CSRShard(..., deterministic=True))")
+        colptr, crow, cval = shard.csc
+        gradp = hiplib.csc_grad_multi(colptr, crow, cval, M, shard.d, kc)
+        if kc != k:
+            return gradp.reshape(shard.d, kc)[:, :k].reshape(-1).contiguous(), lc
+        return gradp, lc
+    z = margins_padded_flat.reshape(n, kc)[:, :k]
+    m, lc = ref_multiplier_multi(z, shard.labels, mask, sample_weight)
+    if not need_grad:
+        return None, lc
+    return ref_csr_grad_multi(shard.rowptr, shard.col, shard.val, m, shard.d), lc
+
+
+def eval_multi_csr(shard, wflat, k, mask=None, need_grad=True,
+                   sample_weight=None):
+    zf = csr_margins_multi(shard, wflat, k)
+    return eval_multi_csr_from_margins(shard, zf, k, mask, need_grad,
+                                       sample_weight)

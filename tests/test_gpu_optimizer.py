@@ -133,6 +133,21 @@ def test_multiclass_large_k_agd_gpu():
     assert acc > 0.3
 
 
+def test_csr_multiclass_agd_gpu():
+    from sparkagd_amd import MultinomialLogisticGradient
+    from sparkagd_amd.data import generate_multiclass_csr_problem
+
+    K = 8
+    shard, _ = generate_multiclass_csr_problem(60000, 50000, 20, num_classes=K,
+                                               seed=33, device=DEV,
+                                               label_noise=0.1)
+    grad = MultinomialLogisticGradient(K)
+    w0 = torch.zeros(50000 * K, device=DEV, dtype=torch.float32)
+    w, h = run(shard, grad, SquaredL2Updater(), 1e-10, 15, 0.001, w0,
+               1.0, math.inf, 0.5, 0.9, True, loss_history_mode="backtrack")
+    assert h[-1] < 0.8 * h[0]
+
+
 def test_agd_on_csr_shard_gpu():
     from sparkagd_amd.data import generate_csr_problem
     from sparkagd_amd import ops

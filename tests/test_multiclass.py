@@ -126,9 +126,54 @@ def test_softmax_trainer_and_model():
                                torch.ones(3000, dtype=proba.dtype))
 
 
-def test_multiclass_guards():
-    from sparkagd_amd.data import generate_csr_problem
+def test_csr_multiclass_matches_dense_oracle():
+    """CSR multiclass margins/grad equal the dense oracle on the densified
+    matrix (CPU tier)."""
+    from sparkagd_amd.data import generate_multiclass_csr_problem
 
+    shard, _ = generate_multiclass_csr_problem(300, 40, 12, num_classes=5, seed=11)
+    K = 5
+    dense = torch.zeros((shard.n, shard.d), dtype=torch.float64)
+    rows = torch.repeat_interleave(
+        torch.arange(shard.n), torch.diff(shard.rowptr.to(torch.int64)))
+    dense.index_put_((rows, shard.col.to(torch.int64)),
+                     shard.val.to(torch.float64), accumulate=True)
+    W = torch.randn(shard.d * K, dtype=torch.float64,
+                    generator=torch.Generator().manual_seed(3))
+    grad_d, lc_d = mc.ref_eval_multi(dense, shard.labels.to(torch.float64), W, K)
+
+    grad = MultinomialLogisticGradient(K)
+    # CSR path computes in f32 (val dtype); compare at f32-appropriate tolerance
+    g_c, lc_c = grad.eval(shard, W)
+    torch.testing.assert_close(g_c.to(torch.float64), grad_d, rtol=2e-5, atol=2e-5)
+    torch.testing.assert_close(lc_c, lc_d, rtol=1e-5, atol=1e-5)
+    # loss-only and from-margins agree
+    zf = grad.margins(shard, W)
+    g2, lc2 = grad.eval_from_margins(shard, zf)
+    torch.testing.assert_close(g2, g_c)
+    torch.testing.assert_close(lc2, lc_c)
+
+
+def test_csr_multiclass_agd_converges():
+    from sparkagd_amd.data import generate_multiclass_csr_problem
+
+    K = 4
+    shard, _ = generate_multiclass_csr_problem(4000, 60, 10, num_classes=K,
+                                               seed=13, label_noise=0.1)
+    grad = MultinomialLogisticGradient(K)
+    w0 = torch.zeros(60 * K, dtype=torch.float32)
+    w, h = run(shard, grad, SquaredL2Updater(), 1e-10, 50, 0.001, w0,
+               1.0, math.inf, 0.5, 0.9, True)
+    assert h[-1] < 0.6 * h[0]
+    # tracked and untracked trajectories agree
+    w_u, h_u = run(shard, grad, SquaredL2Updater(), 1e-10, 50, 0.001, w0,
+                   1.0, math.inf, 0.5, 0.9, True, track_margins=False)
+    assert len(h) == len(h_u)
+    for a, b in zip(h, h_u):
+        assert abs(a - b) < 1e-6 * max(1.0, abs(b))
+
+
+def test_multiclass_guards():
     with pytest.raises(ValueError):
         MultinomialLogisticGradient(1)
     shard, _ = generate_multiclass_problem(100, 5, 3, seed=6, dtype=torch.float64)
@@ -136,6 +181,3 @@ def test_multiclass_guards():
     with pytest.raises(ValueError):
         run(shard, MultinomialLogisticGradient(3), SimpleUpdater(), 1e-6, 3,
             0.0, w0, 1.0, math.inf, 0.5, 0.9, True, solver="gram")
-    csr, _ = generate_csr_problem(50, 10, 3, seed=7)
-    with pytest.raises(ValueError):
-        MultinomialLogisticGradient(3).eval(csr, w0)
