@@ -279,16 +279,27 @@ def eval_multi(features, labels, wflat, k, mask=None, need_grad=True,
 # --- CSR dispatch ---
 
 def csr_margins_multi(shard, wflat: torch.Tensor, k: int) -> torch.Tensor:
-    """Flat padded margins [n*KC] over a CSRShard."""
+    """Flat padded margins [n*KC] over a CSRShard. K > 32 runs the KC=32
+    gather kernel per 32-class chunk (the nnz stream re-reads per chunk —
+    the K-proportional W/M traffic dominates anyway)."""
     kc = padded_k(k)
     if _use_hip(shard.val):
         from . import hiplib
 
-        if kc > 32:
-            raise NotImplementedError(
-                f"CSR multiclass GPU margins support K <= 32 (K={k})")
-        return hiplib.csr_margins_multi(shard.rowptr, shard.col, shard.val,
-                                        wflat, k, kc, shard.d)
+        if kc <= 32:
+            return hiplib.csr_margins_multi(shard.rowptr, shard.col, shard.val,
+                                            wflat, k, kc, shard.d)
+        n, d = shard.n, shard.d
+        w2 = wflat.reshape(d, k).to(torch.float32)
+        z = torch.zeros((n, kc), dtype=torch.float32, device=shard.val.device)
+        for lo in range(0, k, 32):
+            hi = min(lo + 32, k)
+            wc = w2[:, lo:hi]
+            zc = hiplib.csr_margins_multi(shard.rowptr, shard.col, shard.val,
+                                          wc.reshape(-1), hi - lo,
+                                          padded_k(hi - lo), d)
+            z[:, lo:hi] = zc.reshape(n, padded_k(hi - lo))[:, : hi - lo]
+        return z.reshape(-1)
     z = ref_csr_margins_multi(shard.rowptr, shard.col, shard.val, wflat, k,
                               shard.d).reshape(-1, k)
     return _pad_classes(z, kc).reshape(-1)
@@ -301,8 +312,13 @@ def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
     if _use_hip(shard.val):
         from . import hiplib
 
-        M, lc = hiplib.multiplier_multi(margins_padded_flat, shard.labels, k,
-                                        kc, mask, sample_weight)
+        if kc <= 32:
+            M, lc = hiplib.multiplier_multi(margins_padded_flat, shard.labels,
+                                            k, kc, mask, sample_weight)
+        else:  # torch multiplier stage (same route as the dense large-K path)
+            z = margins_padded_flat.reshape(n, kc)[:, :k]
+            m2d, lc = ref_multiplier_multi(z, shard.labels, mask, sample_weight)
+            M = m2d.reshape(-1)
         if not need_grad:
             return None, lc
         if shard.csc is None:
@@ -310,10 +326,24 @@ def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
                 "CSR multiclass gradient needs the deterministic CSC copy "
                 "(CSRShard(..., deterministic=True))")
         colptr, crow, cval = shard.csc
-        gradp = hiplib.csc_grad_multi(colptr, crow, cval, M, shard.d, kc)
-        if kc != k:
-            return gradp.reshape(shard.d, kc)[:, :k].reshape(-1).contiguous(), lc
-        return gradp, lc
+        if kc <= 32:
+            gradp = hiplib.csc_grad_multi(colptr, crow, cval, M, shard.d, kc)
+            if kc != k:
+                return gradp.reshape(shard.d, kc)[:, :k].reshape(-1).contiguous(), lc
+            return gradp, lc
+        m2d = M.reshape(n, k)
+        grad = torch.empty((shard.d, k), dtype=torch.float32,
+                           device=shard.val.device)
+        for lo in range(0, k, 32):
+            hi = min(lo + 32, k)
+            kcc = padded_k(hi - lo)
+            mc_chunk = torch.zeros((n, kcc), dtype=torch.float32,
+                                   device=shard.val.device)
+            mc_chunk[:, : hi - lo] = m2d[:, lo:hi]
+            gc = hiplib.csc_grad_multi(colptr, crow, cval,
+                                       mc_chunk.reshape(-1), shard.d, kcc)
+            grad[:, lo:hi] = gc.reshape(shard.d, kcc)[:, : hi - lo]
+        return grad.reshape(-1).contiguous(), lc
     z = margins_padded_flat.reshape(n, kc)[:, :k]
     m, lc = ref_multiplier_multi(z, shard.labels, mask, sample_weight)
     if not need_grad:

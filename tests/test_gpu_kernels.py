@@ -312,6 +312,31 @@ def test_csr_multiclass_kernels_match_oracle(k):
     torch.testing.assert_close(lh, ln)
 
 
+def test_csr_multiclass_large_k():
+    """K > 32 CSR runs the gather kernels per 32-class chunk."""
+    from sparkagd_amd.data import generate_multiclass_csr_problem
+    from sparkagd_amd.ops import multiclass as mc
+    from sparkagd_amd import MultinomialLogisticGradient
+
+    k = 40
+    shard, _ = generate_multiclass_csr_problem(8000, 10000, 16,
+                                               num_classes=k, seed=35,
+                                               device=DEV)
+    g = torch.Generator(device=DEV).manual_seed(36)
+    W = (torch.randn(shard.d * k, generator=g, device=DEV) /
+         math.sqrt(16)).contiguous()
+    grad = MultinomialLogisticGradient(k)
+    gh, lh = grad.eval(shard, W)
+    zf = mc.ref_csr_margins_multi(shard.rowptr, shard.col, shard.val, W, k,
+                                  shard.d).reshape(-1, k)
+    m, lr = mc.ref_multiplier_multi(zf, shard.labels)
+    gr = mc.ref_csr_grad_multi(shard.rowptr, shard.col, shard.val, m, shard.d)
+    torch.testing.assert_close(gh, gr, rtol=3e-4, atol=3e-4)
+    torch.testing.assert_close(lh, lr, rtol=1e-6, atol=1e-6)
+    gh2, lh2 = grad.eval(shard, W)
+    assert torch.equal(gh, gh2)
+
+
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
 def test_multiclass_large_k(dtype):
     """K > 32 runs GEMM-shaped on the GPU (hipBLASLt / rocBLAS margins+grad,
