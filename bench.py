@@ -258,7 +258,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": args.dtype if device.type == "cuda" else "f32",
+            "dtype": ("f32" if args.csr else args.dtype) if device.type == "cuda" else "f32",
             "data": "synthetic",
             "config": {
                 "model": (f"{'csr_' if args.csr else ''}multinomial{args.classes}_regression"

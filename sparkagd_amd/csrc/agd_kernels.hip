@@ -1436,10 +1436,41 @@ extern "C" int agd_grad_multi(const void* A, int a_dtype, const void* M,
 // the CSC gather keeps the gradient atomics-free => bitwise deterministic.
 // ---------------------------------------------------------------------------
 
-template <int KC>
+// Gather one KC-wide W row into f32 registers. TW=float: KC/4 x 16-B loads.
+// TW=ubf16: half the gathered bytes per nonzero AND double the effective
+// LLC coverage of W (the margins gather is LLC-miss-bound once W [d,KC]
+// exceeds the last-level cache — profiles/r01_csr_multiclass_trace.txt).
+template <typename TW, int KC>
+__device__ __forceinline__ void load_w_row(const TW* __restrict__ wr,
+                                           float (&out)[KC]) {
+  if constexpr (sizeof(TW) == 4) {
+    using f32x4 = __attribute__((ext_vector_type(4))) float;
+#pragma unroll
+    for (int ch = 0; ch < KC / 4; ++ch) {
+      const f32x4 wv = *(const f32x4*)((const float*)wr + ch * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) out[ch * 4 + j] = wv[j];
+    }
+  } else if constexpr (KC >= 8) {  // bf16, 16-B chunks of 8
+    using u16x8 = __attribute__((ext_vector_type(8))) unsigned short;
+#pragma unroll
+    for (int ch = 0; ch < KC / 8; ++ch) {
+      const u16x8 wv = *(const u16x8*)((const ubf16*)wr + ch * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[ch * 8 + j] = bf2f((ubf16)wv[j]);
+    }
+  } else {  // bf16 KC == 4: one 8-B chunk
+    using u16x4 = __attribute__((ext_vector_type(4))) unsigned short;
+    const u16x4 wv = *(const u16x4*)wr;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[j] = bf2f((ubf16)wv[j]);
+  }
+}
+
+template <typename TW, int KC>
 __global__ __launch_bounds__(BLOCK) void k_csr_margins_multi(
     const int* __restrict__ rowptr, const int* __restrict__ col,
-    const float* __restrict__ val, const float* __restrict__ w, ll n,
+    const float* __restrict__ val, const TW* __restrict__ w, ll n,
     float* __restrict__ Z) {
   using f32x4 = __attribute__((ext_vector_type(4))) float;
   const ll stride = (ll)gridDim.x * BLOCK;
@@ -1450,13 +1481,10 @@ __global__ __launch_bounds__(BLOCK) void k_csr_margins_multi(
     for (int j = 0; j < KC; ++j) acc[j] = 0.f;
     for (int k = k_lo; k < k_hi; ++k) {
       const float v = val[k];
-      const float* __restrict__ wr = w + (ll)col[k] * KC;
+      float wv[KC];
+      load_w_row<TW, KC>(w + (ll)col[k] * KC, wv);
 #pragma unroll
-      for (int ch = 0; ch < KC / 4; ++ch) {
-        const f32x4 wv = *(const f32x4*)(wr + ch * 4);
-#pragma unroll
-        for (int j = 0; j < 4; ++j) acc[ch * 4 + j] += v * wv[j];
-      }
+      for (int j = 0; j < KC; ++j) acc[j] += v * wv[j];
     }
     float* __restrict__ zr = Z + r * KC;
 #pragma unroll
@@ -1502,24 +1530,37 @@ __global__ __launch_bounds__(BLOCK) void k_csc_grad_multi(
   }
 }
 
+// w_dtype: 1 = f32 rows, 0 = bf16 rows (half the gather bytes).
 extern "C" int agd_csr_margins_multi(const void* rowptr, const void* col,
                                      const void* val, const void* w,
-                                     long long n, int kc, void* Z,
+                                     long long n, int kc, int w_dtype, void* Z,
                                      void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const int grid = grid_for(n, BLOCK);
-#define LAUNCH_CM(KCV)                                                        \
-  hipLaunchKernelGGL((k_csr_margins_multi<KCV>), dim3(grid), dim3(BLOCK), 0,  \
-                     s, (const int*)rowptr, (const int*)col,                  \
-                     (const float*)val, (const float*)w, n, (float*)Z)
-  switch (kc) {
-    case 4: LAUNCH_CM(4); break;
-    case 8: LAUNCH_CM(8); break;
-    case 16: LAUNCH_CM(16); break;
-    case 32: LAUNCH_CM(32); break;
-    default:
-      snprintf(g_err, sizeof(g_err), "csr_margins_multi: bad KC %d", kc);
-      return 2;
+#define LAUNCH_CM(TW, KCV)                                                    \
+  hipLaunchKernelGGL((k_csr_margins_multi<TW, KCV>), dim3(grid), dim3(BLOCK), \
+                     0, s, (const int*)rowptr, (const int*)col,               \
+                     (const float*)val, (const TW*)w, n, (float*)Z)
+  if (w_dtype == 1) {
+    switch (kc) {
+      case 4: LAUNCH_CM(float, 4); break;
+      case 8: LAUNCH_CM(float, 8); break;
+      case 16: LAUNCH_CM(float, 16); break;
+      case 32: LAUNCH_CM(float, 32); break;
+      default:
+        snprintf(g_err, sizeof(g_err), "csr_margins_multi: bad KC %d", kc);
+        return 2;
+    }
+  } else {
+    switch (kc) {
+      case 4: LAUNCH_CM(ubf16, 4); break;
+      case 8: LAUNCH_CM(ubf16, 8); break;
+      case 16: LAUNCH_CM(ubf16, 16); break;
+      case 32: LAUNCH_CM(ubf16, 32); break;
+      default:
+        snprintf(g_err, sizeof(g_err), "csr_margins_multi: bad KC %d", kc);
+        return 2;
+    }
   }
 #undef LAUNCH_CM
   HIP_CHECK(hipGetLastError());

@@ -79,7 +79,7 @@ def load() -> ctypes.CDLL:
     lib.agd_grad_multi.restype = I
     lib.agd_grad_multi.argtypes = [P, I, P, LL, LL, I, P, LL, P, P]
     lib.agd_csr_margins_multi.restype = I
-    lib.agd_csr_margins_multi.argtypes = [P, P, P, P, LL, I, P, P]
+    lib.agd_csr_margins_multi.argtypes = [P, P, P, P, LL, I, I, P, P]
     lib.agd_csc_grad_multi.restype = I
     lib.agd_csc_grad_multi.argtypes = [P, P, P, P, LL, I, P, P]
 
@@ -449,13 +449,21 @@ def dense_eval_multi_from_margins(
 
 def csr_margins_multi(rowptr, col, val, wflat: torch.Tensor, k: int,
                       kc: int, d: int) -> torch.Tensor:
-    """Padded flat margins [n*KC] = CSR(A) @ pad(W [d,K] -> [d,KC])."""
+    """Padded flat margins [n*KC] = CSR(A) @ pad(W [d,K] -> [d,KC]).
+
+    SPARKAGD_CSR_MULTI_W=bf16 gathers bf16-rounded W rows (half the gather
+    bytes, double the effective LLC coverage of W — the pass is
+    LLC-miss-bound once W exceeds the last-level cache); default f32 keeps
+    exact weights (the CSR values are f32)."""
     lib = load()
     n = rowptr.numel() - 1
     dev = val.device
-    w2 = wflat.reshape(d, k).to(torch.float32)
+    wdt = (torch.bfloat16
+           if os.environ.get("SPARKAGD_CSR_MULTI_W", "f32") == "bf16"
+           else torch.float32)
+    w2 = wflat.reshape(d, k).to(wdt)
     if kc != k:
-        wp = torch.zeros((d, kc), dtype=torch.float32, device=dev)
+        wp = torch.zeros((d, kc), dtype=wdt, device=dev)
         wp[:, :k] = w2
     else:
         wp = w2.contiguous()
@@ -463,6 +471,7 @@ def csr_margins_multi(rowptr, col, val, wflat: torch.Tensor, k: int,
     rc = lib.agd_csr_margins_multi(_ptr(rowptr.contiguous()),
                                    _ptr(col.contiguous()),
                                    _ptr(val.contiguous()), _ptr(wp), n, kc,
+                                   0 if wdt == torch.bfloat16 else 1,
                                    _ptr(Z), _stream(val))
     _check(rc)
     return Z
