@@ -312,6 +312,28 @@ def test_csr_multiclass_kernels_match_oracle(k):
     torch.testing.assert_close(lh, ln)
 
 
+def test_csr_multiclass_bf16_w_gather(monkeypatch):
+    """The optional bf16 W-row gather path matches the oracle with
+    bf16-rounded weights (measured perf-neutral — BACKLOG.md — but kept
+    selectable)."""
+    from sparkagd_amd.data import generate_multiclass_csr_problem
+    from sparkagd_amd.ops import multiclass as mc
+
+    k = 8
+    shard, _ = generate_multiclass_csr_problem(10000, 20000, 16,
+                                               num_classes=k, seed=37,
+                                               device=DEV)
+    g = torch.Generator(device=DEV).manual_seed(38)
+    W = (torch.randn(shard.d * k, generator=g, device=DEV) /
+         math.sqrt(16)).contiguous()
+    Wr = W.reshape(shard.d, k).to(torch.bfloat16).to(torch.float32).reshape(-1)
+    monkeypatch.setenv("SPARKAGD_CSR_MULTI_W", "bf16")
+    zb = mc.csr_margins_multi(shard, W, k)
+    monkeypatch.setenv("SPARKAGD_CSR_MULTI_W", "f32")
+    zr = mc.csr_margins_multi(shard, Wr, k)
+    torch.testing.assert_close(zb, zr, rtol=1e-6, atol=1e-6)
+
+
 def test_csr_multiclass_large_k():
     """K > 32 CSR runs the gather kernels per 32-class chunk."""
     from sparkagd_amd.data import generate_multiclass_csr_problem
