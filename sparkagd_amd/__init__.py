@@ -40,6 +40,7 @@ from .models.trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD
 from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniBatch
 from .gram import GramOperator, run_gram
 from .parallel.comm import Communicator
+from . import evaluation
 
 __version__ = "0.1.0"
 
@@ -63,6 +64,7 @@ __all__ = [
     "LinearModel",
     "MultinomialModel",
     "regularization_path",
+    "evaluation",
     "DenseShard",
     "CSRShard",
     "generate_logistic_data",

@@ -21,13 +21,59 @@ from .gradient import HingeGradient, LeastSquaresGradient, LogisticGradient
 from .updater import SimpleUpdater, SquaredL2Updater, Updater
 
 
+def _save_model(path: str, weights: torch.Tensor, meta: dict) -> None:
+    import json
+    import os as _os
+    import tempfile
+
+    from safetensors.torch import save_file
+
+    d = _os.path.dirname(_os.path.abspath(path)) or "."
+    _os.makedirs(d, exist_ok=True)
+    fd, tmp = tempfile.mkstemp(dir=d, suffix=".tmp")
+    _os.close(fd)
+    try:
+        save_file({"weights": weights.detach().cpu().contiguous().clone()},
+                  tmp, metadata={k: json.dumps(v) for k, v in meta.items()})
+        _os.replace(tmp, path)
+    finally:
+        if _os.path.exists(tmp):
+            _os.unlink(tmp)
+
+
+def _load_model(path: str, device=None):
+    import json
+
+    from safetensors import safe_open
+    from safetensors.torch import load_file
+
+    with safe_open(path, framework="pt", device="cpu") as f:
+        meta = {k: json.loads(v) for k, v in (f.metadata() or {}).items()}
+    w = load_file(path)["weights"]
+    if device is not None:
+        w = w.to(device)
+    return w, meta
+
+
 class LinearModel:
-    """A fitted generalized linear model: weights [d] (+ loss history)."""
+    """A fitted generalized linear model: weights [d] (+ loss history).
+    MLlib-style model persistence via ``save``/``load`` (safetensors)."""
 
     def __init__(self, weights: torch.Tensor, loss_history, link: str):
         self.weights = weights
         self.loss_history = list(loss_history)
         self.link = link
+
+    def save(self, path: str) -> None:
+        _save_model(path, self.weights, {"kind": "linear", "link": self.link,
+                                         "loss_history": self.loss_history})
+
+    @classmethod
+    def load(cls, path: str, device=None) -> "LinearModel":
+        w, meta = _load_model(path, device)
+        if meta.get("kind") != "linear":
+            raise ValueError(f"not a LinearModel checkpoint: {meta.get('kind')}")
+        return cls(w, meta.get("loss_history", []), meta["link"])
 
     def margins(self, features: torch.Tensor) -> torch.Tensor:
         acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
@@ -147,6 +193,18 @@ class MultinomialModel:
 
     def predict_proba(self, features: torch.Tensor) -> torch.Tensor:
         return torch.softmax(self.margins(features), dim=1)
+
+    def save(self, path: str) -> None:
+        _save_model(path, self.weights,
+                    {"kind": "multinomial", "num_classes": self.num_classes,
+                     "loss_history": self.loss_history})
+
+    @classmethod
+    def load(cls, path: str, device=None) -> "MultinomialModel":
+        w, meta = _load_model(path, device)
+        if meta.get("kind") != "multinomial":
+            raise ValueError(f"not a MultinomialModel checkpoint: {meta.get('kind')}")
+        return cls(w, meta.get("loss_history", []), meta["num_classes"])
 
 
 class SoftmaxRegressionWithAGD:

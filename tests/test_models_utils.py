@@ -123,3 +123,57 @@ def test_config_json_roundtrip():
     c.beta = -1
     with pytest.raises(ValueError):
         c.validate()
+
+
+def test_evaluation_metrics():
+    """accuracy/log_loss/roc_auc/precision-recall/confusion vs scikit-learn."""
+    import numpy as np
+    import torch
+    from sklearn.metrics import (roc_auc_score, log_loss as sk_log_loss,
+                                 precision_score, recall_score)
+
+    from sparkagd_amd import evaluation as ev
+
+    rng = np.random.default_rng(4)
+    s = rng.normal(size=3000)
+    y = (rng.random(3000) < 1 / (1 + np.exp(-s))).astype(float)
+    s[:50] = s[0]  # ties exercise the midrank path
+    st, yt = torch.from_numpy(s), torch.from_numpy(y)
+    assert abs(ev.roc_auc(st, yt) - roc_auc_score(y, s)) < 1e-10
+    assert abs(ev.log_loss(st, yt) - sk_log_loss(y, 1 / (1 + np.exp(-s)))) < 1e-9
+    pred = (st > 0).float()
+    prf = ev.precision_recall_f1(pred, yt)
+    assert abs(prf["precision"] - precision_score(y, pred.numpy())) < 1e-12
+    assert abs(prf["recall"] - recall_score(y, pred.numpy())) < 1e-12
+    assert abs(ev.accuracy(pred, yt) - float((pred.numpy() == y).mean())) < 1e-12
+    cm = ev.confusion_matrix(pred, yt, 2)
+    assert int(cm.sum()) == 3000 and int(cm[1, 1]) > 0
+
+
+def test_model_save_load(tmp_path):
+    import math
+    import torch
+
+    from sparkagd_amd import (LogisticRegressionWithAGD, LinearModel,
+                              SoftmaxRegressionWithAGD, MultinomialModel,
+                              generate_dense_problem,
+                              generate_multiclass_problem)
+
+    shard, _ = generate_dense_problem(500, 20, seed=8, dtype=torch.float64)
+    m = LogisticRegressionWithAGD.train(shard, num_iterations=10)
+    p = str(tmp_path / "lin.safetensors")
+    m.save(p)
+    m2 = LinearModel.load(p)
+    assert torch.equal(m.weights.cpu(), m2.weights)
+    assert m2.link == "logistic" and m2.loss_history == m.loss_history
+    assert torch.equal(m.predict(shard.features.cpu()), m2.predict(shard.features.cpu()))
+
+    shard3, _ = generate_multiclass_problem(500, 10, 3, seed=9, dtype=torch.float64)
+    mm = SoftmaxRegressionWithAGD.train(shard3, num_classes=3, num_iterations=10)
+    p3 = str(tmp_path / "soft.safetensors")
+    mm.save(p3)
+    mm2 = MultinomialModel.load(p3)
+    assert torch.equal(mm.weights.cpu(), mm2.weights)
+    assert mm2.num_classes == 3
+    with pytest.raises(ValueError):
+        MultinomialModel.load(p)
