@@ -25,7 +25,14 @@ from sparkagd_amd import (
 from sparkagd_amd.data import DenseShard, shard_range
 
 N = 4000
-PORT = 29781
+def _free_port() -> int:
+    """Ephemeral port reserved by a momentary bind — avoids collisions with
+    fixed ports lingering in TIME_WAIT from earlier runs (flake source)."""
+    import socket
+
+    with socket.socket() as sock:
+        sock.bind(("127.0.0.1", 0))
+        return sock.getsockname()[1]
 
 
 def _worker(rank, world, fn, out_q, port):
@@ -39,7 +46,7 @@ def _worker(rank, world, fn, out_q, port):
         torch.distributed.destroy_process_group()
 
 
-def _run_dist(fn, world=2, port=PORT):
+def _run_dist(fn, world=2, port=_free_port()):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker, args=(r, world, fn, q, port)) for r in range(world)]
@@ -82,7 +89,7 @@ def _dist_minibatch(rank, world):
 def test_sharded_agd_matches_single_process():
     """Row-sharded 2-process AGD == single-process AGD on the same data
     (the replicated-update determinism the design relies on)."""
-    results = _run_dist(_dist_agd, world=2, port=PORT)
+    results = _run_dist(_dist_agd, world=2, port=_free_port())
     full = generate_logistic_data(2.0, -1.5, N, seed=42)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
@@ -100,7 +107,7 @@ def test_sharded_agd_matches_single_process():
 
 
 def test_sharded_minibatch_runs_and_replicates():
-    results = _run_dist(_dist_minibatch, world=2, port=PORT + 1)
+    results = _run_dist(_dist_minibatch, world=2, port=_free_port())
     assert results[0][0] == results[1][0]
     assert len(results[0][1]) == 10
     # loss decreased overall
@@ -123,7 +130,7 @@ def _dist_comm_primitives(rank, world):
 
 
 def test_comm_primitives():
-    results = _run_dist(_dist_comm_primitives, world=2, port=PORT + 2)
+    results = _run_dist(_dist_comm_primitives, world=2, port=_free_port())
     for rank in (0, 1):
         assert all(results[rank]), results[rank]
 
@@ -141,7 +148,7 @@ def _dist_gram_uneven(rank, world):
 
 
 def test_sharded_gram_uneven_rows():
-    results = _run_dist(_dist_gram_uneven, world=2, port=PORT + 5)
+    results = _run_dist(_dist_gram_uneven, world=2, port=_free_port())
     full = generate_logistic_data(2.0, -1.5, N + 1, seed=43)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 5,
@@ -164,7 +171,7 @@ def _dist_gram(rank, world):
 def test_sharded_gram_matches_single_process():
     """2-process Gram solver (cross-rank K blocks via chunked broadcast) ==
     single-process direct solver on the same data."""
-    results = _run_dist(_dist_gram, world=2, port=PORT + 4)
+    results = _run_dist(_dist_gram, world=2, port=_free_port())
     full = generate_logistic_data(2.0, -1.5, N, seed=42)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
@@ -219,7 +226,7 @@ def test_rank_failure_aborts_cleanly_and_resumes(tmp_path):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [
-        ctx.Process(target=_fault_worker, args=(r, 2, ckpt, q, PORT + 3))
+        ctx.Process(target=_fault_worker, args=(r, 2, ckpt, q, _free_port()))
         for r in range(2)
     ]
     for p in procs:
