@@ -46,7 +46,9 @@ def _worker(rank, world, fn, out_q, port):
         torch.distributed.destroy_process_group()
 
 
-def _run_dist(fn, world=2, port=_free_port()):
+def _run_dist(fn, world=2, port=None):
+    if port is None:
+        port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_worker, args=(r, world, fn, q, port)) for r in range(world)]
@@ -225,8 +227,9 @@ def test_rank_failure_aborts_cleanly_and_resumes(tmp_path):
     ckpt = str(tmp_path / "fault.safetensors")
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
+    port = _free_port()
     procs = [
-        ctx.Process(target=_fault_worker, args=(r, 2, ckpt, q, _free_port()))
+        ctx.Process(target=_fault_worker, args=(r, 2, ckpt, q, port))
         for r in range(2)
     ]
     for p in procs:
