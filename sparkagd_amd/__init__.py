@@ -41,6 +41,7 @@ from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniB
 from .gram import GramOperator, run_gram
 from .parallel.comm import Communicator
 from . import evaluation
+from .streaming import HostStreamedDenseShard
 
 __version__ = "0.1.0"
 
@@ -65,6 +66,7 @@ __all__ = [
     "MultinomialModel",
     "regularization_path",
     "evaluation",
+    "HostStreamedDenseShard",
     "DenseShard",
     "CSRShard",
     "generate_logistic_data",
