@@ -109,6 +109,9 @@ def main() -> int:
     p.add_argument("--classes", type=int, default=0,
                    help=">0: multinomial softmax regression with K classes")
     p.add_argument("--eps", type=float, default=1e-3, help="relative loss-improvement epsilon for iters-to-eps")
+    p.add_argument("--streamed", action="store_true",
+                   help="features in pinned host memory, double-buffered H2D "
+                        "streaming (PCIe-bound capacity mode)")
     p.add_argument("--solver", type=str, default="direct", choices=["direct", "gram"],
                    help="gram = dual-space solver (K=A.A^T precompute; O(n_local*n_global) iterations)")
     args = p.parse_args()
@@ -154,6 +157,14 @@ def main() -> int:
             args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
             device=device, dtype=dtype,
         )
+    if args.streamed:
+        if args.classes > 0 or args.csr:
+            raise SystemExit("--streamed supports the dense binary configs")
+        from sparkagd_amd import HostStreamedDenseShard
+
+        shard = HostStreamedDenseShard(
+            shard.features.cpu(), shard.labels,
+            device=device, chunk_rows=min(args.rows, 1024))
     sync(device)
     t_gen = time.perf_counter() - t_gen0
 
@@ -263,7 +274,7 @@ def main() -> int:
             "config": {
                 "model": (f"{'csr_' if args.csr else ''}multinomial{args.classes}_regression"
                           if args.classes > 0
-                          else f"{'csr' if args.csr else 'dense'}_{args.loss}_regression"),
+                          else f"{'csr' if args.csr else ('streamed' if args.streamed else 'dense')}_{args.loss}_regression"),
                 "d": args.d,
                 "rows_per_gpu": args.rows,
                 "global_rows": global_rows,
