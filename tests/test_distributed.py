@@ -161,6 +161,20 @@ def test_sharded_gram_uneven_rows():
             rtol=1e-6, atol=1e-8)
 
 
+def test_sharded_gram_world3():
+    """Odd world size (3 ranks, uneven rows) — exercises the padded
+    all_gather and cross-rank K blocks off the power-of-two path."""
+    results = _run_dist(_dist_gram_uneven, world=3, port=_free_port())
+    full = generate_logistic_data(2.0, -1.5, N + 1, seed=43)
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+    w_ref, _ = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 5,
+                   0.2, w0, 1.0, math.inf, 0.5, 0.9, True)
+    for rank in (0, 1, 2):
+        torch.testing.assert_close(
+            torch.tensor(results[rank][0], dtype=torch.float64), w_ref,
+            rtol=1e-6, atol=1e-8)
+
+
 def _dist_gram(rank, world):
     shard, _ = _make_shard(rank, world)
     comm = Communicator()
