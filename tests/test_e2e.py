@@ -74,3 +74,9 @@ def test_example_scripts_cpu():
     )
     assert out.returncode == 0, out.stderr[-2000:]
     assert "AGD" in out.stdout
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "examples", "train_multiclass.py")],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "save/load round-trip: ok" in out.stdout
