@@ -280,6 +280,7 @@ def main() -> int:
                 "global_rows": global_rows,
                 "parallelism": f"dp{world}",
                 "solver": args.solver,
+                "reg_param": args.reg,
                 "evals_per_step": evals / max(timed_iters, 1),
                 "data_passes_per_step": passes / max(timed_iters, 1),
                 "weights_dtype": str(wdtype).replace("torch.", ""),
