@@ -11,6 +11,11 @@
 //   K6  axpby affine combination    <- Breeze vector ops, AGD.scala:249,255
 //   K7  fused multi-reduction of the 5 per-iteration scalars
 //                                   <- Breeze norm/dot, AGD.scala:263-327
+//   +   multinomial softmax family (beyond the reference, binary-only in
+//       MLlib 1.3): KC-templated VALU margins/multiplier/grad kernels,
+//       hipBLASLt bf16 NT/TN GEMM entries for the dense margins/grad
+//       (skinny GEMMs on the MFMA matrix cores), and CSR gather kernels
+//       with a deterministic CSC-gather transpose.
 //
 // Design notes (MI355X):
 //  * Every data-pass kernel is HBM-bandwidth-bound (arithmetic intensity of a
