@@ -69,6 +69,22 @@ def test_streamed_with_mask_and_weights():
     torch.testing.assert_close(lc1, lc2)
 
 
+def test_streamed_minibatch_matches_dense():
+    """run_mini_batch on a streamed shard == on the in-memory shard (the
+    per-chunk mask slicing path)."""
+    from sparkagd_amd import run_mini_batch
+
+    shard, streamed = _mk(n=1500, d=24)
+    w0 = torch.zeros(24, dtype=torch.float64)
+    w1, h1 = run_mini_batch(shard, LogisticGradient(), SimpleUpdater(),
+                            1.0, 8, 0.0, 0.5, w0)
+    w2, h2 = run_mini_batch(streamed, LogisticGradient(), SimpleUpdater(),
+                            1.0, 8, 0.0, 0.5, w0)
+    torch.testing.assert_close(w1, w2, rtol=1e-12, atol=1e-14)
+    for a, b in zip(h1, h2):
+        assert abs(a - b) < 1e-12 * max(1.0, abs(b))
+
+
 @pytest.mark.gpu
 def test_streamed_gpu_matches_hbm():
     """Streamed (pinned-host + double-buffered H2D) equals the in-HBM shard
