@@ -157,6 +157,10 @@ def main() -> int:
             args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
             device=device, dtype=dtype,
         )
+    if args.solver == "gram" and (args.classes > 0 or args.csr):
+        raise SystemExit("--solver gram supports dense binary shards "
+                         "(multiclass margins are [n,K]; CSR K never fits — "
+                         "see sparkagd_amd/gram.py)")
     if args.streamed:
         if args.classes > 0 or args.csr:
             raise SystemExit("--streamed supports the dense binary configs")
