@@ -230,10 +230,12 @@ def run_gram(
         T += 1
         return t, gm, row
 
-    def finish_basis(t: int, row: torch.Tensor) -> None:
-        row_h = (row / c).cpu().numpy()
+    def finish_basis_host(t: int, row_h: np.ndarray) -> None:
         G[t, : t + 1] = row_h
         G[: t + 1, t] = row_h
+
+    def finish_basis(t: int, row: torch.Tensor) -> None:
+        finish_basis_host(t, (row / c).cpu().numpy())
 
     def quad(a: np.ndarray, b: np.ndarray, k: int) -> float:
         return float(a[:k] @ (G[:k, :k] @ b[:k]))
@@ -310,15 +312,21 @@ def run_gram(
             xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
 
             if beta >= 1.0:
-                f_y = read_loss(lc_y)
-                finish_basis(t_y, row_y)
+                packed = torch.cat([lc_y, row_y]).cpu().numpy()
+                f_y = packed[0] / c
+                finish_basis_host(t_y, packed[2:] / c)
                 break
 
             lc_x = None
             if backtrack_simple:
                 lc_x, _ = eval_loss_async(xm)
-            f_y = read_loss(lc_y)
-            finish_basis(t_y, row_y)
+            # ONE packed D2H transfer for everything this trial must read on
+            # the host (y-loss, the new G row, and the pre-enqueued x-loss)
+            # instead of three small syncs.
+            packed = torch.cat([lc_y, row_y] + ([lc_x] if lc_x is not None
+                                                else [])).cpu().numpy()
+            f_y = packed[0] / c
+            finish_basis_host(t_y, packed[2: 3 + t_y] / c)
 
             k = T + 1
             dxy = cx - cy
@@ -330,7 +338,7 @@ def run_gram(
                 # x == y is excluded above, so reading the pre-enqueued f_x
                 # here matches the reference's compute-f_x-after-the-check
                 # order (and at xy_sq == 0 its value would equal f_y anyway).
-                f_x = read_loss(lc_x)
+                f_x = packed[3 + t_y] / c
                 f_x_bt = f_x
                 xy_dot_gy = float(dxy[:k] @ G[:k, t_y])
                 q_x = f_y + xy_dot_gy + 0.5 * L * xy_sq
