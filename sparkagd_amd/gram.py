@@ -200,9 +200,12 @@ def run_gram(
     G = np.zeros((max_basis + 1, max_basis + 1))
     G[0, 0] = norm_x0_sq
     Mstore = torch.zeros((max_basis, n_local), dtype=acc, device=dev)
-    GMstore = torch.zeros((max_basis, n_local), dtype=torch.float64, device=dev)
+    # XB row 0 = A·x0 margins (f64); row j >= 1 = A·v_j margins of gradient
+    # basis j. Holding them in ONE buffer lets the whole G row for a new
+    # basis vector come from a single dgemv instead of a dgemv + two dots.
+    XB = torch.zeros((max_basis + 1, n_local), dtype=torch.float64, device=dev)
+    XB[0] = xm0.to(torch.float64)
     GMstore32 = torch.zeros((max_basis, n_local), dtype=acc, device=dev)
-    xm0_64 = xm0.to(torch.float64)
     T = 0  # gradient basis vectors so far
 
     def new_basis_async(m_t: torch.Tensor) -> Tuple[int, torch.Tensor, torch.Tensor]:
@@ -218,14 +221,10 @@ def run_gram(
         t = T + 1  # G index
         # G row: dots of v_t with x0 and all previous v_j (fp64, allreduced)
         md = m_t.to(torch.float64)
-        row = torch.empty(t + 1, dtype=torch.float64, device=dev)
-        row[0] = (md * xm0_64).sum()
-        if T > 0:
-            row[1: t] = GMstore[:T] @ md
-        row[t] = (md * gm.to(torch.float64)).sum()
+        XB[t] = gm.to(torch.float64)
+        row = XB[: t + 1] @ md  # dots with x0 and every basis incl. the new one
         comm.allreduce_(row)
         Mstore[T] = m_t
-        GMstore[T] = gm.to(torch.float64)
         GMstore32[T] = gm
         T += 1
         return t, gm, row
