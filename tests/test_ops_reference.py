@@ -81,7 +81,8 @@ def test_dense_eval_mask(loss_type):
 
 def test_csr_eval_matches_dense():
     shard, _ = generate_csr_problem(n=300, d=50, nnz_per_row=5, seed=3)
-    w = torch.randn(50, dtype=torch.float32)
+    w = torch.randn(50, dtype=torch.float32,
+                    generator=torch.Generator().manual_seed(12))
     grad_c, lc_c = shard.eval(w, ops.LOSS_LOGISTIC)
     # densify
     A = torch.zeros(300, 50)
@@ -90,7 +91,8 @@ def test_csr_eval_matches_dense():
             A[i, int(shard.col[k])] += float(shard.val[k])
     grad_d, lc_d = ops.reference.dense_eval(A, shard.labels, w, ops.LOSS_LOGISTIC)
     torch.testing.assert_close(grad_c, grad_d, rtol=1e-4, atol=1e-4)
-    torch.testing.assert_close(lc_c, lc_d, rtol=1e-8, atol=1e-8)
+    # both sides compute in f32 with different summation orders
+    torch.testing.assert_close(lc_c, lc_d, rtol=1e-6, atol=1e-6)
 
 
 def test_csc_build_matches_csr():
