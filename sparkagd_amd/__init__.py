@@ -37,7 +37,8 @@ from .data import (DenseShard, CSRShard, generate_logistic_data,
 from .models.trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD,
                               SVMWithAGD, SoftmaxRegressionWithAGD, LinearModel,
                               MultinomialModel, regularization_path)
-from .optimizer import AcceleratedGradientDescent, run, run_mini_batch, runMiniBatch
+from .optimizer import (AcceleratedGradientDescent, GradientDescent, run,
+                        run_mini_batch, runMiniBatch)
 from .gram import GramOperator, run_gram
 from .parallel.comm import Communicator
 from . import evaluation
@@ -47,6 +48,7 @@ __version__ = "0.1.0"
 
 __all__ = [
     "AGDConfig",
+    "GradientDescent",
     "Gradient",
     "LogisticGradient",
     "LeastSquaresGradient",

@@ -570,3 +570,60 @@ class AcceleratedGradientDescent(Optimizer):
             margin_refresh_every=c.margin_refresh_every,
         )
         return weights
+
+
+class GradientDescent(Optimizer):
+    """Fluent mini-batch SGD optimizer — the analog of MLlib 1.3's
+    ``GradientDescent`` class (the ``Optimizer`` the reference suite's golden
+    baseline ``runMiniBatchSGD`` belongs to, ``Suite.scala:78-86``), with the
+    same setter surface. ``optimize`` delegates to :func:`run_mini_batch`."""
+
+    def __init__(self, gradient: Gradient, updater: Updater,
+                 comm: Optional[Communicator] = None):
+        self.gradient = gradient
+        self.updater = updater
+        self.comm = comm
+        self.step_size = 1.0
+        self.num_iterations = 100
+        self.reg_param = 0.0
+        self.mini_batch_fraction = 1.0
+        self.step_schedule = "sqrt"
+        self.seed = 42
+        self.loss_history: List[float] = []
+
+    def setStepSize(self, step: float) -> "GradientDescent":
+        self.step_size = step
+        return self
+
+    def setNumIterations(self, iters: int) -> "GradientDescent":
+        self.num_iterations = iters
+        return self
+
+    def setRegParam(self, reg_param: float) -> "GradientDescent":
+        self.reg_param = reg_param
+        return self
+
+    def setMiniBatchFraction(self, fraction: float) -> "GradientDescent":
+        self.mini_batch_fraction = fraction
+        return self
+
+    def setStepSchedule(self, schedule: str) -> "GradientDescent":
+        self.step_schedule = schedule
+        return self
+
+    def setGradient(self, gradient: Gradient) -> "GradientDescent":
+        self.gradient = gradient
+        return self
+
+    def setUpdater(self, updater: Updater) -> "GradientDescent":
+        self.updater = updater
+        return self
+
+    def optimize(self, data, initial_weights: torch.Tensor) -> torch.Tensor:
+        weights, self.loss_history = run_mini_batch(
+            data, self.gradient, self.updater, self.step_size,
+            self.num_iterations, self.reg_param, self.mini_batch_fraction,
+            initial_weights, comm=self.comm, seed=self.seed,
+            step_schedule=self.step_schedule,
+        )
+        return weights

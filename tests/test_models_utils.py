@@ -177,3 +177,23 @@ def test_model_save_load(tmp_path):
     assert mm2.num_classes == 3
     with pytest.raises(ValueError):
         MultinomialModel.load(p)
+
+
+def test_gradient_descent_class_api():
+    """Fluent GradientDescent (MLlib's class around runMiniBatchSGD) wires
+    parameters through to run_mini_batch identically."""
+    import torch
+
+    from sparkagd_amd import (GradientDescent, LogisticGradient, SimpleUpdater,
+                              generate_dense_problem, run_mini_batch)
+
+    shard, _ = generate_dense_problem(1500, 12, seed=14, dtype=torch.float64)
+    w0 = torch.zeros(12, dtype=torch.float64)
+    opt = (GradientDescent(LogisticGradient(), SimpleUpdater())
+           .setStepSize(0.8).setNumIterations(9).setMiniBatchFraction(0.5)
+           .setRegParam(0.0).setStepSchedule("sqrt"))
+    w1 = opt.optimize(shard, w0)
+    w2, h2 = run_mini_batch(shard, LogisticGradient(), SimpleUpdater(), 0.8,
+                            9, 0.0, 0.5, w0)
+    torch.testing.assert_close(w1, w2)
+    assert opt.loss_history == h2
