@@ -79,12 +79,23 @@ class LinearModel:
         acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
         return features.to(acc) @ self.weights.to(acc)
 
-    def predict(self, features: torch.Tensor) -> torch.Tensor:
+    def predict(self, features: torch.Tensor,
+                threshold: Optional[float] = 0.5) -> torch.Tensor:
+        """Class predictions. ``threshold`` is the probability cut for
+        logistic models (MLlib's ``setThreshold`` analog; margin cut
+        logit(t)); ``None`` returns the raw margin (``clearThreshold``
+        semantics). Identity-link models always return the raw prediction."""
+        import math as _math
+
         z = self.margins(features)
         if self.link == "identity":
             return z
-        # logistic / hinge: class in {0, 1}
-        return (z > 0).to(torch.float32)
+        if threshold is None:
+            return z
+        cut = 0.0
+        if self.link == "logistic" and threshold != 0.5:
+            cut = _math.log(threshold / (1.0 - threshold))
+        return (z > cut).to(torch.float32)
 
     def predict_proba(self, features: torch.Tensor) -> torch.Tensor:
         if self.link != "logistic":
