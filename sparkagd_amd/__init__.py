@@ -35,7 +35,9 @@ from .models.updater import (
 from .data import (DenseShard, CSRShard, generate_logistic_data,
                    generate_dense_problem, generate_multiclass_problem)
 from .models.trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD,
-                              SVMWithAGD, SoftmaxRegressionWithAGD, LinearModel,
+                              SVMWithAGD, LogisticRegressionWithSGD,
+                              LinearRegressionWithSGD, SVMWithSGD,
+                              SoftmaxRegressionWithAGD, LinearModel,
                               MultinomialModel, regularization_path)
 from .optimizer import (AcceleratedGradientDescent, GradientDescent, run,
                         run_mini_batch, runMiniBatch)
@@ -63,6 +65,9 @@ __all__ = [
     "LogisticRegressionWithAGD",
     "LinearRegressionWithAGD",
     "SVMWithAGD",
+    "LogisticRegressionWithSGD",
+    "LinearRegressionWithSGD",
+    "SVMWithSGD",
     "SoftmaxRegressionWithAGD",
     "LinearModel",
     "MultinomialModel",

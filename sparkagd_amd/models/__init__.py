@@ -12,6 +12,7 @@ from .gradient import (Gradient, LogisticGradient, LeastSquaresGradient,
                        MultinomialLogisticGradient)
 from .updater import Updater, SimpleUpdater, L1Updater, SquaredL2Updater, ElasticNetUpdater
 from .trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD, SVMWithAGD,
+                       LogisticRegressionWithSGD, LinearRegressionWithSGD, SVMWithSGD,
                        SoftmaxRegressionWithAGD, LinearModel, MultinomialModel,
                        regularization_path)
 
@@ -30,6 +31,9 @@ __all__ = [
     "LogisticRegressionWithAGD",
     "LinearRegressionWithAGD",
     "SVMWithAGD",
+    "LogisticRegressionWithSGD",
+    "LinearRegressionWithSGD",
+    "SVMWithSGD",
     "SoftmaxRegressionWithAGD",
     "LinearModel",
     "MultinomialModel",

@@ -253,6 +253,57 @@ class SoftmaxRegressionWithAGD:
         return MultinomialModel(w, opt.loss_history, num_classes)
 
 
+class _SGDTrainer:
+    """MLlib 1.3's primary entry points were ``LogisticRegressionWithSGD``
+    etc. (``GradientDescent``-backed); these are the equivalents, returning
+    the same fitted ``LinearModel`` as the AGD trainers."""
+
+    GRADIENT_CLS = LogisticGradient
+    LINK = "logistic"
+
+    @classmethod
+    def train(
+        cls,
+        data,
+        num_iterations: int = 100,
+        step_size: float = 1.0,
+        mini_batch_fraction: float = 1.0,
+        reg_param: float = 0.0,
+        updater: Optional[Updater] = None,
+        initial_weights: Optional[torch.Tensor] = None,
+        comm: Optional[Communicator] = None,
+        step_schedule: str = "sqrt",
+    ) -> LinearModel:
+        from ..optimizer import GradientDescent
+
+        if updater is None:
+            updater = SquaredL2Updater() if reg_param > 0 else SimpleUpdater()
+        opt = (GradientDescent(cls.GRADIENT_CLS(), updater, comm)
+               .setStepSize(step_size).setNumIterations(num_iterations)
+               .setRegParam(reg_param).setMiniBatchFraction(mini_batch_fraction)
+               .setStepSchedule(step_schedule))
+        if initial_weights is None:
+            wdtype = torch.float64 if data.device.type == "cpu" else torch.float32
+            initial_weights = torch.zeros(data.d, device=data.device, dtype=wdtype)
+        w = opt.optimize(data, initial_weights)
+        return LinearModel(w, opt.loss_history, cls.LINK)
+
+
+class LogisticRegressionWithSGD(_SGDTrainer):
+    GRADIENT_CLS = LogisticGradient
+    LINK = "logistic"
+
+
+class LinearRegressionWithSGD(_SGDTrainer):
+    GRADIENT_CLS = LeastSquaresGradient
+    LINK = "identity"
+
+
+class SVMWithSGD(_SGDTrainer):
+    GRADIENT_CLS = HingeGradient
+    LINK = "hinge"
+
+
 class LogisticRegressionWithAGD(_GLMTrainer):
     GRADIENT_CLS = LogisticGradient
     LINK = "logistic"

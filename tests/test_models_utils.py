@@ -197,3 +197,24 @@ def test_gradient_descent_class_api():
                             9, 0.0, 0.5, w0)
     torch.testing.assert_close(w1, w2)
     assert opt.loss_history == h2
+
+
+def test_sgd_trainers():
+    """The *WithSGD trainer family (MLlib 1.3's primary entry points)."""
+    import torch
+
+    from sparkagd_amd import (LogisticRegressionWithSGD, LinearRegressionWithSGD,
+                              SVMWithSGD, generate_dense_problem, ops)
+
+    shard, _ = generate_dense_problem(2000, 16, seed=15, dtype=torch.float64)
+    m = LogisticRegressionWithSGD.train(shard, num_iterations=25, step_size=1.0)
+    assert m.link == "logistic" and m.loss_history[-1] < m.loss_history[0]
+    acc = float((m.predict(shard.features) == shard.labels).float().mean())
+    assert acc > 0.7
+    shard_l, _ = generate_dense_problem(2000, 16, seed=16, dtype=torch.float64,
+                                        loss_type=ops.LOSS_LEAST_SQUARES)
+    m2 = LinearRegressionWithSGD.train(shard_l, num_iterations=25, step_size=0.3)
+    assert m2.link == "identity" and m2.loss_history[-1] < m2.loss_history[0]
+    m3 = SVMWithSGD.train(shard, num_iterations=25, step_size=0.5,
+                          reg_param=0.01)
+    assert m3.link == "hinge" and m3.loss_history[-1] < m3.loss_history[0]
