@@ -157,9 +157,9 @@ def main() -> int:
             args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
             device=device, dtype=dtype,
         )
-    if args.solver == "gram" and (args.classes > 0 or args.csr):
-        raise SystemExit("--solver gram supports dense binary shards "
-                         "(multiclass margins are [n,K]; CSR K never fits — "
+    if args.solver == "gram" and args.csr:
+        raise SystemExit("--solver gram needs a dense shard (K = A·Aᵀ is "
+                         "dense n×n and the CSR configs have n ≥ 1e6 — "
                          "see sparkagd_amd/gram.py)")
     if args.streamed:
         if args.classes > 0 or args.csr:

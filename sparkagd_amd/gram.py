@@ -124,15 +124,20 @@ class GramOperator:
         self.K = K.contiguous()
         self.build_seconds = time.perf_counter() - t0
 
-    def matvec(self, m_global: torch.Tensor) -> torch.Tensor:
-        """K @ m_global -> local margin slice [n_local] (margins kernel)."""
+    def matvec(self, m_global: torch.Tensor, ncols: int = 1) -> torch.Tensor:
+        """K @ m_global -> local margin slice (flat). ncols > 1 (multiclass:
+        padded class columns) runs one rocBLAS GEMM instead of the GEMV
+        margins kernel; zero pad columns stay exactly zero."""
+        if ncols > 1:
+            mg = m_global.to(self.acc).reshape(self.n_global, ncols)
+            return (self.K @ mg).reshape(-1)
         return ops.dense_margins(self.K, m_global.to(self.acc))
 
-    def all_gather_m(self, m_local: torch.Tensor) -> torch.Tensor:
+    def all_gather_m(self, m_local: torch.Tensor, ncols: int = 1) -> torch.Tensor:
         if self.comm.world_size == 1:
             return m_local
         # NCCL all_gather requires equal lengths: pad to the max count.
-        mx = max(self.counts)
+        mx = max(self.counts) * ncols
         buf = m_local
         if m_local.numel() != mx:
             buf = torch.zeros(mx, dtype=m_local.dtype, device=m_local.device)
@@ -140,7 +145,7 @@ class GramOperator:
         parts = [torch.empty(mx, dtype=m_local.dtype, device=m_local.device)
                  for _ in self.counts]
         dist.all_gather(parts, buf.contiguous())
-        return torch.cat([p[:c] for p, c in zip(parts, self.counts)])
+        return torch.cat([p[: c * ncols] for p, c in zip(parts, self.counts)])
 
 
 def run_gram(
@@ -171,9 +176,16 @@ def run_gram(
         raise ValueError("run_gram requires an affine prox updater (Simple/SquaredL2)")
     if getattr(data, "kind", None) != "dense":
         raise ValueError("run_gram requires a DenseShard")
+    # Multiclass: margins/multipliers are padded [n*KC] flats; every piece of
+    # the coefficient-space machinery below is shape-agnostic over the flat
+    # length (pad columns are exactly zero everywhere so dots are unaffected),
+    # and K applies column-wise (matvec ncols). The basis Gram matrix G holds
+    # Frobenius inner products of the [d,K] basis matrices.
+    ncols = 1
     if getattr(gradient, "IS_MULTICLASS", False):
-        raise ValueError("run_gram does not support multi-class gradients — "
-                         "use the direct solver")
+        from .ops.multiclass import padded_k
+
+        ncols = padded_k(gradient.num_classes)
     backtrack_tol = 1e-10
 
     op = gram_op or GramOperator(data, comm)
@@ -192,20 +204,22 @@ def run_gram(
 
     x0 = initial_weights.clone()
     x0_nonzero = bool(torch.any(x0 != 0))
-    xm0 = gradient.margins(data, x0.to(acc)) if x0_nonzero else torch.zeros(n_local, dtype=acc, device=dev)
+    xm0 = (gradient.margins(data, x0.to(acc)) if x0_nonzero
+           else torch.zeros(n_local * ncols, dtype=acc, device=dev))
     norm_x0_sq = float((x0.to(torch.float64) ** 2).sum())
 
     # basis storage (index 0 = x0; gradients at 1..T)
     max_basis = 8 * num_iterations + 8
     G = np.zeros((max_basis + 1, max_basis + 1))
     G[0, 0] = norm_x0_sq
-    Mstore = torch.zeros((max_basis, n_local), dtype=acc, device=dev)
+    flat_n = n_local * ncols
+    Mstore = torch.zeros((max_basis, flat_n), dtype=acc, device=dev)
     # XB row 0 = A·x0 margins (f64); row j >= 1 = A·v_j margins of gradient
     # basis j. Holding them in ONE buffer lets the whole G row for a new
     # basis vector come from a single dgemv instead of a dgemv + two dots.
-    XB = torch.zeros((max_basis + 1, n_local), dtype=torch.float64, device=dev)
+    XB = torch.zeros((max_basis + 1, flat_n), dtype=torch.float64, device=dev)
     XB[0] = xm0.to(torch.float64)
-    GMstore32 = torch.zeros((max_basis, n_local), dtype=acc, device=dev)
+    GMstore32 = torch.zeros((max_basis, flat_n), dtype=acc, device=dev)
     T = 0  # gradient basis vectors so far
 
     def new_basis_async(m_t: torch.Tensor) -> Tuple[int, torch.Tensor, torch.Tensor]:
@@ -215,8 +229,8 @@ def run_gram(
         nonlocal T
         if T >= max_basis:
             raise RuntimeError("gram basis overflow — raise max_basis")
-        m_global = op.all_gather_m(m_t)
-        gm = op.matvec(m_global)
+        m_global = op.all_gather_m(m_t, ncols)
+        gm = op.matvec(m_global, ncols)
         gm = ops.axpby(1.0 / c, gm, 0.0, gm)  # gm = A·v (local slice)
         t = T + 1  # G index
         # G row: dots of v_t with x0 and all previous v_j (fp64, allreduced)
@@ -405,7 +419,12 @@ def run_gram(
     if T > 0:
         coefs = torch.from_numpy(cx[1: T + 1]).to(device=dev, dtype=acc)
         mcomb = coefs @ Mstore[:T]
-        u = ops.dense_grad_from_mult(feats, mcomb)
+        if ncols > 1:
+            from .ops.multiclass import grad_from_mult_multi
+
+            u = grad_from_mult_multi(feats, mcomb, gradient.num_classes)
+        else:
+            u = ops.dense_grad_from_mult(feats, mcomb)
         comm.allreduce_(u)
         x = (cx[0] * x0.to(acc) + u / c).to(initial_weights.dtype)
     else:

@@ -121,9 +121,9 @@ class MultinomialLogisticGradient(Gradient):
     contiguous); labels are class indices 0..K-1; the loss is the softmax
     cross-entropy; grad = Aᵀ(softmax(Z) − onehot). Works on dense and CSR
     shards (CSR gradients use the deterministic CSC gather).
-    Composes with masks, sample weights and margin-state tracking; the Gram
-    solver does not support the multi-class margin structure (use the
-    direct solver)."""
+    Composes with masks, sample weights, margin-state tracking AND the Gram
+    (dual-space) solver (padded class columns ride the coefficient-space
+    machinery; sparkagd_amd/gram.py)."""
 
     IS_MULTICLASS = True
 
@@ -159,6 +159,15 @@ class MultinomialLogisticGradient(Gradient):
         return mc.eval_multi_from_margins(shard.features, margins, shard.labels,
                                           self.num_classes, mask, need_grad,
                                           shard.sample_weight)
+
+    def multiplier_loss(self, shard, margins, mask=None):
+        """(padded multiplier flat [n*KC], loss_count) from padded margins —
+        the Gram solver's n-space evaluation."""
+        from ..ops import multiclass as mc
+
+        return mc.multiplier_loss_multi(shard.labels, margins,
+                                        self.num_classes, mask,
+                                        getattr(shard, "sample_weight", None))
 
 
 class SmoothedHingeGradient(Gradient):

@@ -356,3 +356,64 @@ def eval_multi_csr(shard, wflat, k, mask=None, need_grad=True,
     zf = csr_margins_multi(shard, wflat, k)
     return eval_multi_csr_from_margins(shard, zf, k, mask, need_grad,
                                        sample_weight)
+
+
+# --- n-space pieces for the Gram (dual-space) solver ---
+
+def multiplier_loss_multi(labels, margins_padded_flat, k, mask=None,
+                          sample_weight=None):
+    """(M padded flat [n*KC] with exact-zero pad columns, loss_count f64[2])
+    from padded margins — zero data passes; the multiclass analog of
+    ops.dense_multiplier_loss for the Gram solver."""
+    kc = padded_k(k)
+    n = margins_padded_flat.numel() // kc
+    if margins_padded_flat.is_cuda and kc <= 32:
+        import os
+
+        if os.environ.get("SPARKAGD_FORCE_REFERENCE") != "1":
+            from . import hiplib
+
+            return hiplib.multiplier_multi(margins_padded_flat, labels, k, kc,
+                                           mask, sample_weight)
+    z = margins_padded_flat.reshape(n, kc)[:, :k]
+    m, lc = ref_multiplier_multi(z, labels, mask, sample_weight)
+    return _pad_classes(m, kc).reshape(-1), lc
+
+
+def grad_from_mult_multi(features, m_padded_flat, k):
+    """grad flat [d*k] = Aᵀ·M from the padded multiplier (dense shards)."""
+    kc = padded_k(k)
+    n, d = features.shape
+    if _use_hip(features):
+        import os
+
+        from . import hiplib
+
+        if (features.dtype == torch.bfloat16
+                and os.environ.get("SPARKAGD_MULTI_GRAD", "auto") != "valu"):
+            gradp = hiplib.gemm_bf16f32_tn(
+                features, m_padded_flat.reshape(n, kc)).reshape(-1)
+        else:
+            if kc > 32:
+                raise NotImplementedError(
+                    "K > 32 VALU grad unsupported — bf16 shards use the GEMM")
+            lib = hiplib.load()
+            n_rb = int(lib.agd_multi_rowblocks(n, d, kc))
+            gradp = torch.empty(d * kc, dtype=torch.float32,
+                                device=features.device)
+            part = (torch.empty(n_rb * d * kc, dtype=torch.float32,
+                                device=features.device)
+                    if n_rb > 1 else gradp)
+            rc = lib.agd_grad_multi(hiplib._ptr(features),
+                                    hiplib._DTYPE_CODE[features.dtype],
+                                    hiplib._ptr(m_padded_flat.contiguous()),
+                                    n, d, kc, hiplib._ptr(part), n_rb,
+                                    hiplib._ptr(gradp),
+                                    hiplib._stream(features))
+            hiplib._check(rc)
+        if kc != k:
+            return gradp.reshape(d, kc)[:, :k].reshape(-1).contiguous()
+        return gradp
+    acc = torch.float32 if features.dtype in (torch.bfloat16, torch.float16) else features.dtype
+    m = m_padded_flat.reshape(n, kc)[:, :k].to(acc)
+    return (features.to(acc).T @ m).reshape(-1)

@@ -198,6 +198,43 @@ def test_sharded_multiclass_matches_single_process():
             assert abs(a - b) < 1e-9 * max(1.0, abs(b))
 
 
+def _dist_gram_multiclass(rank, world):
+    from sparkagd_amd import MultinomialLogisticGradient
+    from sparkagd_amd.data import generate_multiclass_problem
+
+    full, _ = generate_multiclass_problem(901, 10, 3, seed=46,
+                                          dtype=torch.float64, label_noise=0.2)
+    lo, hi = shard_range(901, rank, world)
+    shard = DenseShard(full.features[lo:hi], full.labels[lo:hi])
+    comm = Communicator()
+    w0 = torch.zeros(30, dtype=torch.float64)
+    w, hist = run(shard, MultinomialLogisticGradient(3), SquaredL2Updater(),
+                  1e-12, 8, 0.01, w0, 1.0, math.inf, 0.5, 0.9, True,
+                  comm=comm, solver="gram")
+    return w.numpy().tolist(), hist
+
+
+def test_sharded_gram_multiclass_matches_single():
+    """2-rank Gram solver on a multiclass problem (uneven shards -> padded
+    class-column all_gather) == single-process direct solver."""
+    from sparkagd_amd import MultinomialLogisticGradient
+    from sparkagd_amd.data import generate_multiclass_problem
+
+    results = _run_dist(_dist_gram_multiclass, world=2, port=_free_port())
+    full, _ = generate_multiclass_problem(901, 10, 3, seed=46,
+                                          dtype=torch.float64, label_noise=0.2)
+    w0 = torch.zeros(30, dtype=torch.float64)
+    w_ref, hist_ref = run(full, MultinomialLogisticGradient(3),
+                          SquaredL2Updater(), 1e-12, 8, 0.01, w0,
+                          1.0, math.inf, 0.5, 0.9, True)
+    for rank in (0, 1):
+        torch.testing.assert_close(
+            torch.tensor(results[rank][0], dtype=torch.float64), w_ref,
+            rtol=1e-6, atol=1e-8)
+        for a, b in zip(results[rank][1], hist_ref):
+            assert abs(a - b) < 1e-7 * max(1.0, abs(b))
+
+
 def test_sharded_gram_world3():
     """Odd world size (3 ranks, uneven rows) — exercises the padded
     all_gather and cross-rank K blocks off the power-of-two path."""

@@ -173,11 +173,36 @@ def test_csr_multiclass_agd_converges():
         assert abs(a - b) < 1e-6 * max(1.0, abs(b))
 
 
+def test_multiclass_gram_matches_direct():
+    """The dual-space (Gram) solver on a multiclass problem follows the
+    direct solver's trajectory (padded class columns through the
+    coefficient-space machinery)."""
+    K = 5
+    shard, _ = generate_multiclass_problem(400, 24, K, seed=17,
+                                           dtype=torch.float64,
+                                           label_noise=0.2)
+    grad = MultinomialLogisticGradient(K)
+    w0 = torch.zeros(24 * K, dtype=torch.float64)
+    args = (grad, SquaredL2Updater(), 1e-12, 25, 0.01, w0,
+            1.0, math.inf, 0.5, 0.9, True)
+    w_d, h_d = run(shard, *args, loss_history_mode="backtrack")
+    w_g, h_g = run(shard, *args, loss_history_mode="backtrack", solver="gram")
+    assert len(h_d) == len(h_g)
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 1e-8 * max(1.0, abs(b)), (a, b)
+    torch.testing.assert_close(w_g, w_d, rtol=1e-6, atol=1e-9)
+    # nonzero initial weights exercise the x0 margin basis row
+    w1 = torch.randn(24 * K, dtype=torch.float64,
+                     generator=torch.Generator().manual_seed(18)) * 0.05
+    w_d2, h_d2 = run(shard, grad, SquaredL2Updater(), 1e-12, 10, 0.01, w1,
+                     1.0, math.inf, 0.5, 0.9, True)
+    w_g2, h_g2 = run(shard, grad, SquaredL2Updater(), 1e-12, 10, 0.01, w1,
+                     1.0, math.inf, 0.5, 0.9, True, solver="gram")
+    for a, b in zip(h_d2, h_g2):
+        assert abs(a - b) < 1e-8 * max(1.0, abs(b))
+    torch.testing.assert_close(w_g2, w_d2, rtol=1e-6, atol=1e-9)
+
+
 def test_multiclass_guards():
     with pytest.raises(ValueError):
         MultinomialLogisticGradient(1)
-    shard, _ = generate_multiclass_problem(100, 5, 3, seed=6, dtype=torch.float64)
-    w0 = torch.zeros(15, dtype=torch.float64)
-    with pytest.raises(ValueError):
-        run(shard, MultinomialLogisticGradient(3), SimpleUpdater(), 1e-6, 3,
-            0.0, w0, 1.0, math.inf, 0.5, 0.9, True, solver="gram")
