@@ -88,6 +88,10 @@ class CountingGradient:
     def IS_MULTICLASS(self):
         return getattr(self.inner, "IS_MULTICLASS", False)
 
+    @property
+    def num_classes(self):
+        return self.inner.num_classes
+
 
 def sync(device: torch.device) -> None:
     if device.type == "cuda":
