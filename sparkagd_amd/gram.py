@@ -26,7 +26,10 @@ instead of ~66 GB of shard traffic. The trade-offs, stated plainly:
   absolute time-to-solution still wins whenever n_global ≲ d;
 * requires a dense shard, an AFFINE prox (Simple/SquaredL2), full-batch
   evaluations (no mini-batch masks), and fp32 (or fp64) accumulation
-  identical in class to the direct path.
+  identical in class to the direct path. Multiclass (softmax) gradients are
+  supported: margins/multipliers are padded [n·KC] flats whose zero pad
+  columns leave every inner product unchanged, and K applies column-wise
+  (one rocBLAS GEMM per trial).
 
 The trajectory is the same mathematics as the direct solver up to fp
 rounding (asserted by tests/test_gram.py against the direct path).
