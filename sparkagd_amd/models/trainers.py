@@ -156,6 +156,7 @@ def regularization_path(
 
     gradient = gradient or LogisticGradient()
     updater = SquaredL2Updater()
+    is_multi = getattr(gradient, "IS_MULTICLASS", False)
     link = {0: "logistic", 1: "identity", 2: "hinge", 3: "hinge"}.get(
         gradient.LOSS_TYPE, "logistic")
     comm = comm or Communicator()
@@ -168,7 +169,8 @@ def regularization_path(
                 raise
             op = None
     wdtype = torch.float64 if data.device.type == "cpu" else torch.float32
-    w = torch.zeros(data.d, device=data.device, dtype=wdtype)
+    dim = data.d * gradient.num_classes if is_multi else data.d
+    w = torch.zeros(dim, device=data.device, dtype=wdtype)
     models = []
     for lam in lambdas:
         w0 = w if warm_start else torch.zeros_like(w)
@@ -180,7 +182,11 @@ def regularization_path(
             loss_history_mode="backtrack", comm=comm,
             solver="gram" if op is not None else "direct", gram_op=op,
         )
-        models.append(LinearModel(w.clone(), hist, link))
+        if is_multi:
+            models.append(MultinomialModel(w.clone(), hist,
+                                           gradient.num_classes))
+        else:
+            models.append(LinearModel(w.clone(), hist, link))
     return models
 
 

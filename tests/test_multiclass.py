@@ -206,3 +206,21 @@ def test_multiclass_gram_matches_direct():
 def test_multiclass_guards():
     with pytest.raises(ValueError):
         MultinomialLogisticGradient(1)
+
+
+def test_multiclass_regularization_path():
+    """regularization_path with a multiclass gradient returns
+    MultinomialModels and reuses the Gram operator across lambdas."""
+    from sparkagd_amd import MultinomialModel, regularization_path
+
+    shard, _ = generate_multiclass_problem(600, 15, 4, seed=19,
+                                           dtype=torch.float64,
+                                           label_noise=0.2)
+    models = regularization_path(shard, [0.1, 0.01, 0.001],
+                                 gradient=MultinomialLogisticGradient(4),
+                                 num_iterations=25)
+    assert len(models) == 3
+    assert all(isinstance(m, MultinomialModel) for m in models)
+    acc = float((models[-1].predict(shard.features) == shard.labels)
+                .float().mean())
+    assert acc > 0.7
