@@ -170,6 +170,7 @@ def run_gram(
     metrics=None,
     iteration_hook=None,
     gram_op: Optional[GramOperator] = None,
+    backtrack_tol: float = 1e-10,
 ) -> Tuple[torch.Tensor, List[float]]:
     """AGD in coefficient space over the gradient basis (same 12-parameter
     surface as optimizer.run; same AT/backtracking/restart state machine;
@@ -189,7 +190,6 @@ def run_gram(
         from .ops.multiclass import padded_k
 
         ncols = padded_k(gradient.num_classes)
-    backtrack_tol = 1e-10
 
     op = gram_op or GramOperator(data, comm)
     acc = op.acc
@@ -222,7 +222,6 @@ def run_gram(
     # basis vector come from a single dgemv instead of a dgemv + two dots.
     XB = torch.zeros((max_basis + 1, flat_n), dtype=torch.float64, device=dev)
     XB[0] = xm0.to(torch.float64)
-    GMstore32 = torch.zeros((max_basis, flat_n), dtype=acc, device=dev)
     T = 0  # gradient basis vectors so far
 
     def new_basis_async(m_t: torch.Tensor) -> Tuple[int, torch.Tensor, torch.Tensor]:
@@ -242,7 +241,6 @@ def run_gram(
         row = XB[: t + 1] @ md  # dots with x0 and every basis incl. the new one
         comm.allreduce_(row)
         Mstore[T] = m_t
-        GMstore32[T] = gm
         T += 1
         return t, gm, row
 

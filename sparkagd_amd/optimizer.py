@@ -111,6 +111,7 @@ def run(
     margin_refresh_every: int = 0,
     solver: str = "direct",
     gram_op=None,
+    backtrack_tol: float = 1e-10,
 ) -> Tuple[torch.Tensor, List[float]]:
     """Run accelerated proximal gradient descent.
 
@@ -156,10 +157,10 @@ def run(
             reg_param, initial_weights, L0, Lexact, beta, alpha, may_restart,
             loss_history_mode=loss_history_mode, comm=comm, metrics=metrics,
             iteration_hook=iteration_hook, gram_op=gram_op,
+            backtrack_tol=backtrack_tol,
         )
 
     comm = comm or Communicator()
-    backtrack_tol = 1e-10
 
     x = initial_weights.clone()
     z = x.clone()
@@ -568,6 +569,7 @@ class AcceleratedGradientDescent(Optimizer):
             solver=c.solver,
             track_margins=c.track_margins,
             margin_refresh_every=c.margin_refresh_every,
+            backtrack_tol=c.backtrack_tol,
         )
         return weights
 

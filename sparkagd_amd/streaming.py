@@ -99,8 +99,13 @@ class HostStreamedDenseShard:
             b = i % 2
             buf = self._buf[b][: hi - lo]
             with torch.cuda.stream(self._copy_stream):
-                if i >= 2:  # buffer reuse: wait for compute of chunk i-2
-                    self._copy_stream.wait_event(self._compute_done[b])
+                # Buffer reuse: wait for the LAST compute that read this
+                # buffer — chunk i-2 of this pass, or (for i < 2) the tail
+                # chunks of the PREVIOUS pass, whose events persist across
+                # calls. Waiting unconditionally covers back-to-back passes
+                # with no intervening host sync (an event never recorded is
+                # a no-op wait).
+                self._copy_stream.wait_event(self._compute_done[b])
                 buf.copy_(self.features_host[lo:hi], non_blocking=True)
                 self._copy_done[b].record(self._copy_stream)
             main.wait_event(self._copy_done[b])

@@ -25,20 +25,23 @@ from sparkagd_amd import (
 from sparkagd_amd.data import DenseShard, shard_range
 
 N = 4000
-def _free_port() -> int:
-    """Ephemeral port reserved by a momentary bind — avoids collisions with
-    fixed ports lingering in TIME_WAIT from earlier runs (flake source)."""
-    import socket
-
-    with socket.socket() as sock:
-        sock.bind(("127.0.0.1", 0))
-        return sock.getsockname()[1]
 
 
-def _worker(rank, world, fn, out_q, port):
-    os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(port)
-    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+def _init_file() -> str:
+    """Fresh FileStore rendezvous path. File-based init removes the
+    bind-release-rebind race an ephemeral TCP port helper has (another
+    process can grab the port between release and rebind)."""
+    import tempfile
+
+    fd, path = tempfile.mkstemp(prefix="sparkagd_dist_", suffix=".init")
+    os.close(fd)
+    os.unlink(path)  # init_method="file://" wants a nonexistent path
+    return path
+
+
+def _worker(rank, world, fn, out_q, init_file):
+    torch.distributed.init_process_group(
+        "gloo", init_method=f"file://{init_file}", rank=rank, world_size=world)
     try:
         res = fn(rank, world)
         out_q.put((rank, res))
@@ -47,21 +50,25 @@ def _worker(rank, world, fn, out_q, port):
 
 
 def _run_dist(fn, world=2, port=None):
-    if port is None:
-        port = _free_port()
+    init_file = _init_file()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, world, fn, q, port)) for r in range(world)]
+    procs = [ctx.Process(target=_worker, args=(r, world, fn, q, init_file))
+             for r in range(world)]
     for p in procs:
         p.start()
-    results = {}
-    for _ in range(world):
-        rank, res = q.get(timeout=300)
-        results[rank] = res
-    for p in procs:
-        p.join(timeout=60)
-        assert p.exitcode == 0
-    return results
+    try:
+        results = {}
+        for _ in range(world):
+            rank, res = q.get(timeout=300)
+            results[rank] = res
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        return results
+    finally:
+        if os.path.exists(init_file):
+            os.unlink(init_file)
 
 
 def _make_shard(rank, world):
@@ -91,7 +98,7 @@ def _dist_minibatch(rank, world):
 def test_sharded_agd_matches_single_process():
     """Row-sharded 2-process AGD == single-process AGD on the same data
     (the replicated-update determinism the design relies on)."""
-    results = _run_dist(_dist_agd, world=2, port=_free_port())
+    results = _run_dist(_dist_agd, world=2)
     full = generate_logistic_data(2.0, -1.5, N, seed=42)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
@@ -109,7 +116,7 @@ def test_sharded_agd_matches_single_process():
 
 
 def test_sharded_minibatch_runs_and_replicates():
-    results = _run_dist(_dist_minibatch, world=2, port=_free_port())
+    results = _run_dist(_dist_minibatch, world=2)
     assert results[0][0] == results[1][0]
     assert len(results[0][1]) == 10
     # loss decreased overall
@@ -132,7 +139,7 @@ def _dist_comm_primitives(rank, world):
 
 
 def test_comm_primitives():
-    results = _run_dist(_dist_comm_primitives, world=2, port=_free_port())
+    results = _run_dist(_dist_comm_primitives, world=2)
     for rank in (0, 1):
         assert all(results[rank]), results[rank]
 
@@ -150,7 +157,7 @@ def _dist_gram_uneven(rank, world):
 
 
 def test_sharded_gram_uneven_rows():
-    results = _run_dist(_dist_gram_uneven, world=2, port=_free_port())
+    results = _run_dist(_dist_gram_uneven, world=2)
     full = generate_logistic_data(2.0, -1.5, N + 1, seed=43)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 5,
@@ -183,7 +190,7 @@ def test_sharded_multiclass_matches_single_process():
     from sparkagd_amd import MultinomialLogisticGradient
     from sparkagd_amd.data import generate_multiclass_problem
 
-    results = _run_dist(_dist_multiclass, world=2, port=_free_port())
+    results = _run_dist(_dist_multiclass, world=2)
     full, _ = generate_multiclass_problem(1200, 12, 4, seed=44,
                                           dtype=torch.float64, label_noise=0.1)
     w0 = torch.zeros(48, dtype=torch.float64)
@@ -220,7 +227,7 @@ def test_sharded_gram_multiclass_matches_single():
     from sparkagd_amd import MultinomialLogisticGradient
     from sparkagd_amd.data import generate_multiclass_problem
 
-    results = _run_dist(_dist_gram_multiclass, world=2, port=_free_port())
+    results = _run_dist(_dist_gram_multiclass, world=2)
     full, _ = generate_multiclass_problem(901, 10, 3, seed=46,
                                           dtype=torch.float64, label_noise=0.2)
     w0 = torch.zeros(30, dtype=torch.float64)
@@ -238,7 +245,7 @@ def test_sharded_gram_multiclass_matches_single():
 def test_sharded_gram_world3():
     """Odd world size (3 ranks, uneven rows) — exercises the padded
     all_gather and cross-rank K blocks off the power-of-two path."""
-    results = _run_dist(_dist_gram_uneven, world=3, port=_free_port())
+    results = _run_dist(_dist_gram_uneven, world=3)
     full = generate_logistic_data(2.0, -1.5, N + 1, seed=43)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, _ = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 5,
@@ -261,7 +268,7 @@ def _dist_gram(rank, world):
 def test_sharded_gram_matches_single_process():
     """2-process Gram solver (cross-rank K blocks via chunked broadcast) ==
     single-process direct solver on the same data."""
-    results = _run_dist(_dist_gram, world=2, port=_free_port())
+    results = _run_dist(_dist_gram, world=2)
     full = generate_logistic_data(2.0, -1.5, N, seed=42)
     w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
     w_ref, hist_ref = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
@@ -284,13 +291,11 @@ def test_sharded_gram_matches_single_process():
 # an uninterrupted run at the same iteration count.
 # ---------------------------------------------------------------------------
 
-def _fault_worker(rank, world, ckpt_path, out_q, port):
+def _fault_worker(rank, world, ckpt_path, out_q, init_file):
     import datetime
 
-    os.environ["MASTER_ADDR"] = "127.0.0.1"
-    os.environ["MASTER_PORT"] = str(port)
     torch.distributed.init_process_group(
-        "gloo", rank=rank, world_size=world,
+        "gloo", init_method=f"file://{init_file}", rank=rank, world_size=world,
         timeout=datetime.timedelta(seconds=20),
     )
     shard, _ = _make_shard(rank, world)
@@ -315,9 +320,9 @@ def test_rank_failure_aborts_cleanly_and_resumes(tmp_path):
     ckpt = str(tmp_path / "fault.safetensors")
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = _free_port()
+    init_file = _init_file()
     procs = [
-        ctx.Process(target=_fault_worker, args=(r, 2, ckpt, q, port))
+        ctx.Process(target=_fault_worker, args=(r, 2, ckpt, q, init_file))
         for r in range(2)
     ]
     for p in procs:
@@ -347,3 +352,63 @@ def test_rank_failure_aborts_cleanly_and_resumes(tmp_path):
                           0.2, w0, 1.0, math.inf, 0.5, 0.9, True)
     torch.testing.assert_close(w_res, w_ref, rtol=1e-9, atol=1e-12)
     assert len(hist_res) == len(hist_ref)
+
+
+# ---------------------------------------------------------------------------
+# init_from_env: the torchrun rendezvous path bench.py runs under
+# (RANK/WORLD_SIZE/MASTER_* env vars -> init_process_group -> Communicator).
+# Exercised here with the gloo backend; on a GPU node the same code selects
+# nccl (=RCCL) and binds the device from LOCAL_RANK.
+# ---------------------------------------------------------------------------
+
+def _env_worker(rank, world, port, out_q):
+    from sparkagd_amd.parallel.comm import init_from_env
+
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    comm = init_from_env()
+    try:
+        t = torch.full((4,), float(rank + 1), dtype=torch.float64)
+        comm.allreduce_(t)
+        out_q.put((rank, (comm.rank, comm.world_size, t.numpy().tolist())))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_init_from_env_rendezvous():
+    import socket
+
+    ctx = mp.get_context("spawn")
+    last_err = None
+    for _attempt in range(3):  # env:// needs a real port; retry on collision
+        with socket.socket() as sock:
+            sock.bind(("127.0.0.1", 0))
+            port = sock.getsockname()[1]
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_env_worker, args=(r, 2, port, q))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        try:
+            results = {}
+            for _ in range(2):
+                rank, res = q.get(timeout=120)
+                results[rank] = res
+            for p in procs:
+                p.join(timeout=60)
+                assert p.exitcode == 0
+            for rank in (0, 1):
+                got_rank, got_world, reduced = results[rank]
+                assert got_rank == rank and got_world == 2
+                assert reduced == [3.0, 3.0, 3.0, 3.0]
+            return
+        except Exception as e:  # noqa: BLE001 - port stolen between probes
+            last_err = e
+            for p in procs:
+                if p.is_alive():
+                    p.terminate()
+                p.join(timeout=30)
+    raise AssertionError(f"init_from_env rendezvous failed 3x: {last_err}")
