@@ -34,14 +34,9 @@ if dev.type == "cuda":
     torch.cuda.synchronize()
 t_gd = time.perf_counter() - t0
 
+from sparkagd_amd.utils.metrics import iters_to_eps  # noqa: E402 (shared definition with bench.py)
+
 lstar = min(min(h_agd), min(h_gd))
-
-
-def iters_to(hist, target):
-    for i, v in enumerate(hist):
-        if v <= target:
-            return i + 1
-    return None
 
 
 print(f"config: dense logistic d={d} n={n} {'bf16' if dev.type=='cuda' else 'f64'}, L2 {REG}")
@@ -50,6 +45,6 @@ print(f"GD : {len(h_gd)} iters in {t_gd:.2f}s ({1e3*t_gd/len(h_gd):.1f} ms/it), 
 print(f"L* = {lstar:.8f}")
 print(f"{'eps':>8} {'AGD iters':>10} {'GD iters':>10}")
 for eps in (0.5, 0.2, 0.1, 0.05, 0.02, 0.01, 0.001):
-    ta = iters_to(h_agd, lstar * (1 + eps) + 1e-15)
-    tg = iters_to(h_gd, lstar * (1 + eps) + 1e-15)
+    ta = iters_to_eps(h_agd, eps, loss_star=lstar)
+    tg = iters_to_eps(h_gd, eps, loss_star=lstar)
     print(f"{eps:>8} {str(ta):>10} {str(tg):>10}")

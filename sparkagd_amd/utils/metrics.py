@@ -50,3 +50,24 @@ class JsonlMetrics:
 
     def __exit__(self, *a) -> None:
         self.close()
+
+
+def iters_to_eps(history, eps: float, loss_star: Optional[float] = None):
+    """Iterations to ε-accuracy: the first (1-based) iteration whose recorded
+    objective is ≤ (1 + eps)·L*, the distance-to-optimum definition of the
+    BASELINE metric's second half (the at-scale version of the reference's
+    AGD(10) ≈ GD(50) iteration-advantage contract, ``Suite.scala:60-90``).
+
+    ``loss_star`` defaults to min(history) — callers should pass a history
+    long enough to have converged (bench.py extends the run past the timed
+    window for exactly this). Returns None if the target is never reached.
+    The single shared definition for bench.py and benchmarks/iters_to_eps.py.
+    """
+    if not history:
+        return None
+    lstar = min(history) if loss_star is None else loss_star
+    target = lstar * (1.0 + eps) + 1e-15  # absolute guard for L* == 0
+    for i, v in enumerate(history):
+        if v <= target:
+            return i + 1
+    return None
