@@ -32,7 +32,7 @@ from .models.updater import (
     SquaredL2Updater,
     ElasticNetUpdater,
 )
-from .data import (DenseShard, CSRShard, generate_logistic_data,
+from .data import (DenseShard, CSRShard, MixedShard, generate_logistic_data,
                    generate_dense_problem, generate_multiclass_problem)
 from .models.trainers import (LogisticRegressionWithAGD, LinearRegressionWithAGD,
                               SVMWithAGD, LogisticRegressionWithSGD,
@@ -76,6 +76,7 @@ __all__ = [
     "HostStreamedDenseShard",
     "DenseShard",
     "CSRShard",
+    "MixedShard",
     "generate_logistic_data",
     "generate_dense_problem",
     "generate_multiclass_problem",

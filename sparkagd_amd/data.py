@@ -359,3 +359,100 @@ def generate_csr_problem(
     else:
         labels = (z + noise > 0).to(torch.float32)
     return CSRShard(rowptr, col, val, labels, d), w_true
+
+
+class MixedShard:
+    """Heterogeneous shard: dense AND CSR-sparse example blocks in one
+    logical row space.
+
+    MLlib's ``Gradient.compute`` accepts dense or sparse vectors per example
+    within one RDD (invoked at ``AGD.scala:198``); the MI355X-native analog
+    groups examples by representation — a preprocessing pass sorts rows into
+    one dense block and one sparse block per rank (order does not matter for
+    full-batch sums) — and evaluates each block with its own fused kernels,
+    summing the (grad, loss, count) partials on device. Margins concatenate
+    in block order, so margin-state tracking and mini-batch masks compose
+    exactly as for homogeneous shards.
+
+    Binary losses only (the multiclass gradient dispatches on homogeneous
+    shard kinds). All parts must share the feature dimension and device.
+    """
+
+    kind = "mixed"
+
+    def __init__(self, parts):
+        parts = list(parts)
+        if not parts:
+            raise ValueError("MixedShard needs at least one part")
+        d0, dev0 = parts[0].d, parts[0].device
+        for p in parts:
+            if p.d != d0:
+                raise ValueError(f"feature-dim mismatch: {p.d} != {d0}")
+            if p.device != dev0:
+                raise ValueError("all parts must live on one device")
+        self.parts = parts
+        self._offsets = [0]
+        for p in parts:
+            self._offsets.append(self._offsets[-1] + p.n)
+        self.labels = torch.cat([p.labels.to(parts[0].labels.dtype)
+                                 for p in parts])
+        sw = [getattr(p, "sample_weight", None) for p in parts]
+        self.sample_weight = None
+        if any(w is not None for w in sw):
+            self.sample_weight = torch.cat([
+                w if w is not None
+                else torch.ones(p.n, dtype=torch.float32, device=dev0)
+                for w, p in zip(sw, parts)])
+
+    @property
+    def n(self) -> int:
+        return self._offsets[-1]
+
+    @property
+    def d(self) -> int:
+        return self.parts[0].d
+
+    @property
+    def device(self) -> torch.device:
+        return self.parts[0].device
+
+    @property
+    def nbytes(self) -> int:
+        return sum(p.nbytes for p in self.parts)
+
+    def _split(self, t: Optional[torch.Tensor]):
+        if t is None:
+            return [None] * len(self.parts)
+        return [t[lo:hi].contiguous() for lo, hi in
+                zip(self._offsets[:-1], self._offsets[1:])]
+
+    def eval(self, w: torch.Tensor, loss_type: int,
+             mask: Optional[torch.Tensor] = None, need_grad: bool = True):
+        grad = None
+        loss_count = None
+        for p, m in zip(self.parts, self._split(mask)):
+            g, lc = p.eval(w, loss_type, m, need_grad)
+            loss_count = lc if loss_count is None else loss_count + lc
+            if need_grad:
+                grad = g if grad is None else grad + g.to(grad.dtype)
+        return grad, loss_count
+
+    def margins(self, v: torch.Tensor) -> torch.Tensor:
+        ms = [p.margins(v) for p in self.parts]
+        dt = ms[0].dtype
+        for m in ms[1:]:
+            dt = torch.promote_types(dt, m.dtype)
+        return torch.cat([m.to(dt) for m in ms])
+
+    def eval_from_margins(self, margins: torch.Tensor, loss_type: int,
+                          mask: Optional[torch.Tensor] = None,
+                          need_grad: bool = True):
+        grad = None
+        loss_count = None
+        for p, vm, m in zip(self.parts, self._split(margins),
+                            self._split(mask)):
+            g, lc = p.eval_from_margins(vm, loss_type, m, need_grad)
+            loss_count = lc if loss_count is None else loss_count + lc
+            if need_grad:
+                grad = g if grad is None else grad + g.to(grad.dtype)
+        return grad, loss_count

@@ -109,3 +109,20 @@ def test_gram_operator_reuse():
     w2, h2 = run_gram(data, LogisticGradient(), SimpleUpdater(), 1e-12, 5, 0.0,
                       w0, 1.0, math.inf, 0.5, 0.9, True, gram_op=op)
     assert torch.equal(w1, w2) and h1 == h2
+
+
+def test_gram_checkpoint_resume_explicit_error():
+    """Checkpoint/resume is a direct-solver capability; the Gram solver must
+    refuse it LOUDLY (VERDICT r01 #6), not silently skip checkpointing."""
+    import pytest
+
+    full = generate_logistic_data(2.0, -1.5, 500, seed=9)
+    w0 = torch.tensor([0.1, 0.1], dtype=torch.float64)
+    with pytest.raises(ValueError, match="direct solver"):
+        run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 3, 0.1, w0,
+            1.0, math.inf, 0.5, 0.9, True, solver="gram",
+            checkpoint_path="/tmp/never_written.safetensors", checkpoint_every=1)
+    with pytest.raises(ValueError, match="direct solver"):
+        run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 3, 0.1, w0,
+            1.0, math.inf, 0.5, 0.9, True, solver="gram",
+            resume_from="/tmp/does_not_exist.safetensors")

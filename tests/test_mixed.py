@@ -1,0 +1,117 @@
+"""MixedShard: heterogeneous dense + CSR example blocks in one shard.
+
+MLlib accepts dense or sparse vectors per example within one RDD
+(``AGD.scala:198`` invokes ``Gradient.compute`` on whatever representation
+each example carries); here examples are grouped by representation into
+blocks evaluated by their own fused kernels (VERDICT r01 'Missing #3').
+"""
+
+import math
+
+import torch
+
+from sparkagd_amd import (
+    LogisticGradient,
+    SimpleUpdater,
+    SquaredL2Updater,
+    run,
+    run_mini_batch,
+)
+from sparkagd_amd.data import CSRShard, DenseShard, MixedShard, generate_csr_problem
+
+
+def _split_mixed(n_dense=1200, n=2400, d=300, nnz=16, seed=31):
+    """One CSR problem split into a densified head block + CSR tail block,
+    plus the all-dense equivalent for parity."""
+    full, _ = generate_csr_problem(n, d, nnz, seed=seed)
+    dense_all = torch.zeros((n, d), dtype=torch.float32)
+    rows = torch.repeat_interleave(torch.arange(n),
+                                   torch.diff(full.rowptr.to(torch.int64)))
+    # index_put_ with accumulate: generate_csr_problem can emit duplicate
+    # (row, col) pairs, which fancy-index += would silently drop
+    dense_all.index_put_((rows, full.col.to(torch.int64)), full.val,
+                         accumulate=True)
+    head = DenseShard(dense_all[:n_dense].clone(), full.labels[:n_dense])
+    lo = int(full.rowptr[n_dense])
+    tail = CSRShard(full.rowptr[n_dense:].to(torch.int64) - lo,
+                    full.col[lo:], full.val[lo:],
+                    full.labels[n_dense:], d)
+    mixed = MixedShard([head, tail])
+    ref = DenseShard(dense_all, full.labels)
+    return mixed, ref
+
+
+def test_mixed_eval_matches_dense():
+    mixed, ref = _split_mixed()
+    torch.manual_seed(3)
+    w = torch.randn(300, dtype=torch.float32) * 0.1
+    g_m, lc_m = mixed.eval(w, LogisticGradient.LOSS_TYPE)
+    g_r, lc_r = ref.eval(w, LogisticGradient.LOSS_TYPE)
+    assert float(lc_m[1]) == float(lc_r[1]) == 2400.0
+    assert abs(float(lc_m[0]) - float(lc_r[0])) < 1e-5 * abs(float(lc_r[0]))
+    torch.testing.assert_close(g_m.float(), g_r.float(), rtol=1e-4, atol=1e-5)
+
+
+def test_mixed_agd_matches_dense_trajectory():
+    """Full AGD (with margin-state tracking) over the mixed shard equals the
+    all-dense run of the same rows — block order only changes fp summation
+    order."""
+    mixed, ref = _split_mixed()
+    w0 = torch.zeros(300, dtype=torch.float32)
+    args = (LogisticGradient(), SquaredL2Updater(), 1e-12, 8, 0.05)
+    w_m, h_m = run(mixed, *args, w0, 1.0, math.inf, 0.5, 0.9, True)
+    w_r, h_r = run(ref, *args, w0, 1.0, math.inf, 0.5, 0.9, True)
+    assert len(h_m) == len(h_r)
+    for a, b in zip(h_m, h_r):
+        assert abs(a - b) < 1e-4 * max(1.0, abs(b)), (a, b)
+    torch.testing.assert_close(w_m, w_r, rtol=1e-3, atol=1e-5)
+
+
+def test_mixed_minibatch_mask_split():
+    """Seeded Bernoulli masks split correctly across the blocks (the mask is
+    drawn over the mixed shard's whole row space)."""
+    mixed, ref = _split_mixed()
+    w0 = torch.zeros(300, dtype=torch.float32)
+    w_m, h_m = run_mini_batch(mixed, LogisticGradient(), SimpleUpdater(), 1.0,
+                              6, 0.0, 0.5, w0, seed=11)
+    w_r, h_r = run_mini_batch(ref, LogisticGradient(), SimpleUpdater(), 1.0,
+                              6, 0.0, 0.5, w0, seed=11)
+    assert len(h_m) == len(h_r) == 6
+    for a, b in zip(h_m, h_r):
+        assert abs(a - b) < 1e-4 * max(1.0, abs(b)), (a, b)
+
+
+def test_mixed_shard_validation():
+    import pytest
+
+    d1 = DenseShard(torch.randn(10, 5), torch.zeros(10))
+    d2 = DenseShard(torch.randn(10, 6), torch.zeros(10))
+    with pytest.raises(ValueError, match="feature-dim"):
+        MixedShard([d1, d2])
+    with pytest.raises(ValueError, match="at least one"):
+        MixedShard([])
+    m = MixedShard([d1])
+    assert m.n == 10 and m.d == 5 and m.nbytes == d1.nbytes
+
+
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_mixed_agd_gpu():
+    """Mixed dense+CSR blocks through the HIP kernels end to end."""
+    mixed, ref = _split_mixed()
+    dev = "cuda:0"
+    gm = MixedShard([
+        DenseShard(mixed.parts[0].features.to(dev), mixed.parts[0].labels.to(dev)),
+        CSRShard(mixed.parts[1].rowptr.to(dev), mixed.parts[1].col.to(dev),
+                 mixed.parts[1].val.to(dev), mixed.parts[1].labels.to(dev),
+                 mixed.parts[1].d),
+    ])
+    w0 = torch.zeros(300, dtype=torch.float32, device=dev)
+    args = (LogisticGradient(), SquaredL2Updater(), 1e-12, 8, 0.05)
+    w_g, h_g = run(gm, *args, w0, 1.0, math.inf, 0.5, 0.9, True)
+    w_c, h_c = run(mixed, *args, w0.cpu(), 1.0, math.inf, 0.5, 0.9, True)
+    assert len(h_g) == len(h_c)
+    for a, b in zip(h_g, h_c):
+        assert abs(a - b) < 1e-3 * max(1.0, abs(b)), (a, b)

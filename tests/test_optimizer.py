@@ -225,3 +225,60 @@ def test_restart_fires():
                         1.0, math.inf, 0.5, 0.9, False)
     # Restart should not be (much) worse; on this problem it typically helps.
     assert hist_r[-1] <= hist_nr[-1] * 1.5
+
+
+def test_backtrack_tol_config_is_honored():
+    """config.backtrack_tol must actually reach the backtracking state
+    machine (advisor finding r01: it was silently ignored). Observable: with
+    tol=0 the |f_y - f_x| >= tol test always holds, so the solver stays on
+    the SIMPLE test (loss-only f_x evaluations: need_grad=False); with
+    tol=inf it switches to the alternate test after one trial (full-gradient
+    f_x evaluations: need_grad=True)."""
+    import math as _math
+
+    data = generate_logistic_data(2.0, -1.5, 2000, seed=55)
+
+    class CountingGradient(LogisticGradient):
+        def __init__(self):
+            self.loss_only = 0
+            self.with_grad = 0
+
+        def eval(self, shard, w, mask=None, need_grad=True):
+            if need_grad:
+                self.with_grad += 1
+            else:
+                self.loss_only += 1
+            return super().eval(shard, w, mask, need_grad)
+
+        def eval_from_margins(self, shard, margins, mask=None, need_grad=True):
+            if need_grad:
+                self.with_grad += 1
+            else:
+                self.loss_only += 1
+            return super().eval_from_margins(shard, margins, mask, need_grad)
+
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+
+    g0 = CountingGradient()
+    run(data, g0, SimpleUpdater(), 1e-12, 8, 0.0, w0, 1.0, _math.inf,
+        0.5, 0.9, True, backtrack_tol=0.0, loss_history_mode="backtrack")
+    g_inf = CountingGradient()
+    run(data, g_inf, SimpleUpdater(), 1e-12, 8, 0.0, w0, 1.0, _math.inf,
+        0.5, 0.9, True, backtrack_tol=_math.inf, loss_history_mode="backtrack")
+    # tol=0: every backtracking trial is loss-only (simple test forever)
+    assert g0.loss_only >= 8
+    # tol=inf: at most the very first trial is loss-only, the rest carry grads
+    assert g_inf.loss_only <= 1
+    assert g_inf.with_grad > g0.with_grad
+
+    # and the class API threads it from the config
+    from sparkagd_amd import AcceleratedGradientDescent, AGDConfig
+
+    g_cfg = CountingGradient()
+    opt = AcceleratedGradientDescent(
+        g_cfg, SimpleUpdater(),
+        config=AGDConfig(num_iterations=8, convergence_tol=1e-12,
+                         backtrack_tol=_math.inf,
+                         loss_history_mode="backtrack"))
+    opt.optimize(data, w0)
+    assert g_cfg.loss_only <= 1
