@@ -98,8 +98,12 @@ class CSRShard:
     def __init__(self, rowptr: torch.Tensor, col: torch.Tensor, val: torch.Tensor,
                  labels: torch.Tensor, d: int, deterministic: bool = True,
                  sample_weight: Optional[torch.Tensor] = None):
-        self.rowptr = rowptr.contiguous()
-        self.col = col.contiguous()
+        # canonical int32 indices (the HIP kernels' index width); guard the
+        # ranges instead of silently truncating
+        if val.numel() > 2**31 - 1 or d > 2**31 - 1:
+            raise ValueError("CSR shard exceeds int32 index range — split it")
+        self.rowptr = rowptr.contiguous().to(torch.int32)
+        self.col = col.contiguous().to(torch.int32)
         self.val = val.contiguous()
         self.labels = labels.to(device=val.device)
         if self.labels.dtype not in (torch.float32, torch.float64):

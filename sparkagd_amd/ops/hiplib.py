@@ -468,8 +468,8 @@ def csr_margins_multi(rowptr, col, val, wflat: torch.Tensor, k: int,
     else:
         wp = w2.contiguous()
     Z = torch.empty(n * kc, dtype=torch.float32, device=dev)
-    rc = lib.agd_csr_margins_multi(_ptr(rowptr.contiguous()),
-                                   _ptr(col.contiguous()),
+    rowptr, col = _csr_idx(rowptr, col)
+    rc = lib.agd_csr_margins_multi(_ptr(rowptr), _ptr(col),
                                    _ptr(val.contiguous()), _ptr(wp), n, kc,
                                    0 if wdt == torch.bfloat16 else 1,
                                    _ptr(Z), _stream(val))
@@ -527,12 +527,26 @@ def gemm_bf16f32_tn(A: torch.Tensor, M: torch.Tensor) -> torch.Tensor:
     return grad
 
 
+def _csr_idx(rowptr: torch.Tensor, col: torch.Tensor):
+    """Kernel index arrays are int32: convert int64 inputs (no-op otherwise).
+    Passing int64 straight through would silently misread (the bug the
+    MixedShard GPU test caught in round 2)."""
+    rowptr = rowptr.contiguous()
+    col = col.contiguous()
+    if rowptr.dtype != torch.int32:
+        rowptr = rowptr.to(torch.int32)
+    if col.dtype != torch.int32:
+        col = col.to(torch.int32)
+    return rowptr, col
+
+
 def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
     lib = load()
     n = rowptr.numel() - 1
+    rowptr, col = _csr_idx(rowptr, col)
     margins = torch.empty(n, dtype=torch.float32, device=val.device)
     rc = lib.agd_csr_eval(
-        _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
+        _ptr(rowptr), _ptr(col), _ptr(val.contiguous()),
         None, None, None, _ptr(v.contiguous()), n, val.numel(), v.numel(),
         None, None, _ptr(margins), None, 0, None, None, None, 0, 1,
         _ptr(_red_ws(val.device)), _stream(val),
@@ -546,6 +560,7 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
                           sample_weight=None):
     lib = load()
     n = rowptr.numel() - 1
+    rowptr, col = _csr_idx(rowptr, col)
     d = d if d is not None else 0
     labels = labels.contiguous()
     if labels.dtype != torch.float32:
@@ -562,7 +577,7 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
     mult = torch.empty(n, dtype=torch.float32, device=dev)
     sw = _prep_weights(sample_weight, val.device)
     rc = lib.agd_csr_eval(
-        _ptr(rowptr.contiguous()), _ptr(col.contiguous()), _ptr(val.contiguous()),
+        _ptr(rowptr), _ptr(col), _ptr(val.contiguous()),
         _ptr(labels), _ptr(mask), _ptr(sw), None, n, val.numel(), d,
         _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
         loss_type, _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 2,

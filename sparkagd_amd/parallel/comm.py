@@ -105,8 +105,12 @@ def init_from_env(backend: Optional[str] = None) -> Communicator:
         return Communicator()
     if not dist.is_initialized():
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
-        if backend == "nccl" and torch.cuda.is_available():
+            # SPARKAGD_DIST_BACKEND=gloo lets a multi-rank run share one GPU
+            # (RCCL rejects duplicate devices in a communicator — measured,
+            # profiles/r02_rccl_2rank_probe.txt); default is nccl(=RCCL).
+            backend = os.environ.get("SPARKAGD_DIST_BACKEND") or (
+                "nccl" if torch.cuda.is_available() else "gloo")
+        if torch.cuda.is_available():
             local_rank = int(os.environ.get("LOCAL_RANK", os.environ["RANK"]))
             torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
