@@ -112,7 +112,13 @@ def main() -> int:
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--rows", type=int, default=16384, help="rows per GPU (weak scaling)")
-    p.add_argument("--d", type=int, default=1_000_000)
+    # --dim is an alias for --d: torchrun's own argparse abbreviation-matches
+    # a bare --d before the script args when launching under
+    # torch.distributed.run, so multi-rank launches must use --dim
+    p.add_argument("--d", "--dim", dest="d", type=int, default=1_000_000)
+    p.add_argument("--label-noise", type=float, default=0.1,
+                   help="planted-model label noise; ~1.0 makes the problem "
+                        "non-separable (persistent hinge-active rows)")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "f32", "f8"])
     p.add_argument("--loss", type=str, default="logistic", choices=list(LOSSES))
     # Default L2 reg 1e-3: with n << d any synthetic labeling is linearly
@@ -177,7 +183,7 @@ def main() -> int:
     else:
         shard, _w_true = generate_dense_problem(
             args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
-            device=device, dtype=dtype,
+            device=device, dtype=dtype, label_noise=args.label_noise,
         )
     if args.solver == "gram" and args.csr:
         raise SystemExit("--solver gram needs a dense shard (K = A·Aᵀ is "
@@ -331,6 +337,7 @@ def main() -> int:
                 "parallelism": f"dp{world}",
                 "solver": args.solver,
                 "reg_param": args.reg,
+                "label_noise": args.label_noise,
                 "evals_per_step": evals / max(timed_iters, 1),
                 "data_passes_per_step": passes / max(timed_iters, 1),
                 "examples_definition": "rows x loss_evaluations (2 evals/AGD step); rows_per_sec = rows x steps / s",
