@@ -84,6 +84,12 @@ __device__ __forceinline__ T wave_reduce_sum(T v) {
   return v;
 }
 
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
 // Block-level fp64 reduction of NACC accumulators into a per-block partial
 // slot; a one-block k_reduce_partials finishes the sum. Two stages instead
 // of leader atomics: 8192 same-word fp64 atomics serialize to ~0.5 ms
@@ -659,6 +665,61 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier_multi(
     const float loss = (logf(esum) + zmax) - ((y >= 0 && y < K) ? z[y] : 0.f);
     lsum += (double)loss * (double)scale;
     cnt += (double)scale;
+  }
+  double acc[2] = {lsum, cnt};
+  block_reduce_partial<2>(acc, red_part);
+}
+
+// Generic-K multiplier (K > 32, any 4-aligned KC): one WAVE per row, lanes
+// stride the class axis, so no per-thread K-sized register array is needed.
+// z is streamed twice (max+expsum pass, then the write pass recomputes the
+// exps) — still ONE launch and 3 n*KC streams, vs the ~6-kernel torch stage
+// this replaces (logsumexp/gather/softmax/scatter_add/mul/sum); the n*K
+// space is tiny next to the A streams either way, the win is launch/driver
+// overhead off the hot loop and exact-zero pad columns for the GEMM grad.
+__global__ __launch_bounds__(BLOCK) void k_multiplier_multi_anyk(
+    const float* __restrict__ Z, const float* __restrict__ labels,
+    const unsigned char* __restrict__ mask,
+    const float* __restrict__ sample_weight, ll n, int K, int KC,
+    float* __restrict__ M, double* __restrict__ red_part) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
+  const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
+  double lsum = 0.0, cnt = 0.0;
+  for (ll i = wave_gid; i < n; i += n_waves) {
+    const float* __restrict__ zrow = Z + i * KC;
+    float* __restrict__ mrow = M + i * KC;
+    if (mask && !mask[i]) {
+      for (int k = lane; k < KC; k += WAVE) mrow[k] = 0.f;
+      continue;
+    }
+    const int y = (int)labels[i];
+    float zmax = -3.0e38f;
+    for (int k = lane; k < K; k += WAVE) zmax = fmaxf(zmax, zrow[k]);
+    zmax = wave_reduce_max(zmax);
+    float esum = 0.f;
+    for (int k = lane; k < K; k += WAVE) esum += __expf(zrow[k] - zmax);
+    esum = wave_reduce_sum(esum);
+    const float inv = 1.0f / esum;
+    const float scale = sample_weight ? sample_weight[i] : 1.0f;
+    float zy = 0.f;  // only the lane iteration with k == y contributes
+    for (int k = lane; k < KC; k += WAVE) {
+      if (k < K) {
+        const float zk = zrow[k];
+        const float m = __expf(zk - zmax) * inv - ((k == y) ? 1.0f : 0.0f);
+        mrow[k] = m * scale;
+        if (k == y) zy = zk;
+      } else {
+        mrow[k] = 0.f;
+      }
+    }
+    zy = wave_reduce_sum(zy);
+    if (lane == 0) {
+      const float loss = (logf(esum) + zmax) - ((y >= 0 && y < K) ? zy : 0.f);
+      lsum += (double)loss * (double)scale;
+      cnt += (double)scale;
+    }
   }
   double acc[2] = {lsum, cnt};
   block_reduce_partial<2>(acc, red_part);
@@ -1377,18 +1438,30 @@ extern "C" int agd_multiplier_multi(const void* Z, const void* labels,
                      s, (const float*)Z, (const float*)labels,                \
                      (const unsigned char*)mask, (const float*)sample_weight, \
                      n, k, (float*)M, (double*)red_ws)
+  int launched_grid = grid;
   switch (kc) {
     case 4: LAUNCH_MU(4); break;
     case 8: LAUNCH_MU(8); break;
     case 16: LAUNCH_MU(16); break;
     case 32: LAUNCH_MU(32); break;
-    default:
-      snprintf(g_err, sizeof(g_err), "multiplier_multi: bad KC %d", kc);
-      return 2;
+    default: {
+      if (kc < 4 || (kc & 3)) {
+        snprintf(g_err, sizeof(g_err), "multiplier_multi: bad KC %d", kc);
+        return 2;
+      }
+      // generic K > 32: wave per row, lanes stride classes
+      launched_grid = grid_for(n, WAVES_PER_BLOCK);
+      hipLaunchKernelGGL(k_multiplier_multi_anyk, dim3(launched_grid),
+                         dim3(BLOCK), 0, s, (const float*)Z,
+                         (const float*)labels, (const unsigned char*)mask,
+                         (const float*)sample_weight, n, k, kc, (float*)M,
+                         (double*)red_ws);
+      break;
+    }
   }
 #undef LAUNCH_MU
   hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, s,
-                     (double*)red_ws, grid, (double*)loss_count);
+                     (double*)red_ws, launched_grid, (double*)loss_count);
   HIP_CHECK(hipGetLastError());
   return 0;
 }

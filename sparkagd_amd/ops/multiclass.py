@@ -208,42 +208,26 @@ def margins_multi(features: torch.Tensor, wflat: torch.Tensor, k: int) -> torch.
 
 def _eval_large_k_from_margins(features, margins_padded_flat, labels, k, kc,
                                mask, need_grad, sample_weight):
-    """K > 32 multiplier stage in plain torch on the GPU (softmax over n×K is
-    tiny next to the A streams), grad as the hipBLASLt TN GEMM (bf16 shards)
-    or rocBLAS torch.mm (f32 shards)."""
+    """K > 32 path: the generic-K multiplier kernel (wave-per-row
+    k_multiplier_multi_anyk — ONE launch instead of the ~6-kernel torch
+    stage this replaced in round 1), then grad as the hipBLASLt TN GEMM
+    (bf16 shards, padded M fed straight in — its pad columns are exact
+    zeros) or rocBLAS torch.mm (f32 shards)."""
     import os
 
     from . import hiplib
 
     n, d = features.shape
-    z = margins_padded_flat.reshape(n, kc)[:, :k]
-    y = labels.to(torch.int64)
-    lse = torch.logsumexp(z, dim=1)
-    loss = lse - z.gather(1, y.unsqueeze(1)).squeeze(1)
-    m = torch.softmax(z, dim=1)
-    m = m.scatter_add(1, y.unsqueeze(1),
-                      -torch.ones((n, 1), dtype=z.dtype, device=z.device))
-    count_t = None
-    if mask is not None or sample_weight is not None:
-        scale = torch.ones(n, dtype=z.dtype, device=z.device)
-        if mask is not None:
-            scale = scale * mask.to(z.dtype)
-        if sample_weight is not None:
-            scale = scale * sample_weight.to(z.dtype)
-        loss = loss * scale
-        m = m * scale.unsqueeze(1)
-        count_t = scale.to(torch.float64).sum()
-    if count_t is None:
-        count_t = torch.tensor(float(n), dtype=torch.float64, device=z.device)
-    loss_count = torch.stack([loss.to(torch.float64).sum(), count_t])
+    M, loss_count = hiplib.multiplier_multi(margins_padded_flat, labels, k, kc,
+                                            mask, sample_weight)
     if not need_grad:
         return None, loss_count
     if (features.dtype == torch.bfloat16
             and os.environ.get("SPARKAGD_MULTI_GRAD", "auto") != "valu"):
-        mp = torch.zeros((n, kc), dtype=torch.float32, device=z.device)
-        mp[:, :k] = m
-        return hiplib.gemm_bf16f32_tn(features, mp).reshape(d, kc)[:, :k].reshape(-1).contiguous(), loss_count
+        grad = hiplib.gemm_bf16f32_tn(features, M.reshape(n, kc))
+        return grad.reshape(d, kc)[:, :k].reshape(-1).contiguous(), loss_count
     if features.dtype == torch.float32:
+        m = M.reshape(n, kc)[:, :k]
         return (features.T @ m).reshape(-1), loss_count
     raise NotImplementedError(
         f"K={k} > 32 gradient needs a bf16 (hipBLASLt) or f32 (rocBLAS) shard")
@@ -312,13 +296,10 @@ def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
     if _use_hip(shard.val):
         from . import hiplib
 
-        if kc <= 32:
-            M, lc = hiplib.multiplier_multi(margins_padded_flat, shard.labels,
-                                            k, kc, mask, sample_weight)
-        else:  # torch multiplier stage (same route as the dense large-K path)
-            z = margins_padded_flat.reshape(n, kc)[:, :k]
-            m2d, lc = ref_multiplier_multi(z, shard.labels, mask, sample_weight)
-            M = m2d.reshape(-1)
+        # the multiplier kernel covers every KC (templated <= 32, generic
+        # wave-per-row above)
+        M, lc = hiplib.multiplier_multi(margins_padded_flat, shard.labels,
+                                        k, kc, mask, sample_weight)
         if not need_grad:
             return None, lc
         if shard.csc is None:
@@ -331,7 +312,7 @@ def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
             if kc != k:
                 return gradp.reshape(shard.d, kc)[:, :k].reshape(-1).contiguous(), lc
             return gradp, lc
-        m2d = M.reshape(n, k)
+        m2d = M.reshape(n, kc)[:, :k]
         grad = torch.empty((shard.d, k), dtype=torch.float32,
                            device=shard.val.device)
         for lo in range(0, k, 32):
@@ -367,7 +348,7 @@ def multiplier_loss_multi(labels, margins_padded_flat, k, mask=None,
     ops.dense_multiplier_loss for the Gram solver."""
     kc = padded_k(k)
     n = margins_padded_flat.numel() // kc
-    if margins_padded_flat.is_cuda and kc <= 32:
+    if margins_padded_flat.is_cuda:
         import os
 
         if os.environ.get("SPARKAGD_FORCE_REFERENCE") != "1":
