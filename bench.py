@@ -130,6 +130,13 @@ def main() -> int:
     p.add_argument("--reg", type=float, default=1e-3)
     p.add_argument("--csr", action="store_true", help="CSR-sparse shard instead of dense")
     p.add_argument("--nnz-per-row", type=int, default=64)
+    p.add_argument("--csr-dist", type=str, default="uniform",
+                   choices=["uniform", "zipf"],
+                   help="column popularity: uniform, or zipf (power-law)")
+    p.add_argument("--csr-zipf-a", type=float, default=1.1)
+    p.add_argument("--csr-cluster", action="store_true",
+                   help="column-frequency clustering preprocessing "
+                        "(reindex_columns) before training")
     p.add_argument("--classes", type=int, default=0,
                    help=">0: multinomial softmax regression with K classes")
     p.add_argument("--eps", type=float, default=1e-3,
@@ -178,8 +185,13 @@ def main() -> int:
     elif args.csr:
         shard, _w_true = generate_csr_problem(
             args.rows, args.d, args.nnz_per_row, seed=1234 + rank * 7,
-            loss_type=loss_type, device=device,
+            loss_type=loss_type, device=device, col_dist=args.csr_dist,
+            zipf_a=args.csr_zipf_a,
         )
+        if args.csr_cluster:
+            from sparkagd_amd.data import reindex_columns
+
+            shard, _perm = reindex_columns(shard)
     else:
         shard, _w_true = generate_dense_problem(
             args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
@@ -338,6 +350,8 @@ def main() -> int:
                 "solver": args.solver,
                 "reg_param": args.reg,
                 "label_noise": args.label_noise,
+                "csr_dist": args.csr_dist if args.csr else None,
+                "csr_cluster": bool(args.csr_cluster) if args.csr else None,
                 "evals_per_step": evals / max(timed_iters, 1),
                 "data_passes_per_step": passes / max(timed_iters, 1),
                 "examples_definition": "rows x loss_evaluations (2 evals/AGD step); rows_per_sec = rows x steps / s",

@@ -337,14 +337,31 @@ def generate_csr_problem(
     seed: int,
     loss_type: int = ops.LOSS_LOGISTIC,
     device: str | torch.device = "cpu",
+    col_dist: str = "uniform",
+    zipf_a: float = 1.1,
 ) -> Tuple[CSRShard, torch.Tensor]:
-    """Random CSR problem: each row has ``nnz_per_row`` uniformly drawn
-    (sorted, possibly duplicate-free not guaranteed) column indices with
-    N(0,1) values; planted labels as in :func:`generate_dense_problem`."""
+    """Random CSR problem: each row has ``nnz_per_row`` column indices drawn
+    uniformly or Zipf-distributed (``col_dist="zipf"``: column popularity
+    ~ rank^-a with hot columns scattered over the ID space by a seeded
+    permutation — the realistic power-law regime that column-frequency
+    clustering, :func:`reindex_columns`, targets), with N(0,1) values;
+    planted labels as in :func:`generate_dense_problem`."""
     dev = torch.device(device)
     gen = torch.Generator(device=dev).manual_seed(seed)
     nnz = n * nnz_per_row
-    col = torch.randint(0, d, (nnz,), generator=gen, device=dev, dtype=torch.int32)
+    if col_dist == "zipf":
+        ranks = torch.arange(1, d + 1, device=dev, dtype=torch.float64)
+        cdf = ranks.pow_(-float(zipf_a)).cumsum_(0)
+        cdf /= cdf[-1].clone()
+        u = torch.rand(nnz, generator=gen, device=dev, dtype=torch.float64)
+        rank_idx = torch.searchsorted(cdf, u).clamp_(max=d - 1)
+        scatter = torch.randperm(d, generator=gen, device=dev)
+        col = scatter[rank_idx].to(torch.int32)
+        del ranks, cdf, u, rank_idx, scatter
+    elif col_dist == "uniform":
+        col = torch.randint(0, d, (nnz,), generator=gen, device=dev, dtype=torch.int32)
+    else:
+        raise ValueError("col_dist must be uniform|zipf")
     col = col.view(n, nnz_per_row).sort(dim=1).values.reshape(-1).contiguous()
     val = torch.randn(nnz, generator=gen, device=dev, dtype=torch.float32)
     rowptr = torch.arange(0, nnz + 1, nnz_per_row, device=dev, dtype=torch.int32)
@@ -460,3 +477,52 @@ class MixedShard:
             if need_grad:
                 grad = g if grad is None else grad + g.to(grad.dtype)
         return grad, loss_count
+
+
+def reindex_columns(shard: CSRShard):
+    """Column-frequency clustering preprocessing (round-2 backlog item):
+    renumber feature columns by descending occurrence count so the hottest
+    columns occupy the lowest IDs — their weight entries then pack into a
+    handful of cache lines that stay LLC/L2-resident, and a wave gathering a
+    row's w[col] entries coalesces several lanes onto the same 64 B line
+    (the CSR margins pass is gather-line-service-bound at d=1e7,
+    profiles/r01_csr_kernel_trace.txt).
+
+    Returns ``(clustered_shard, perm)`` where ``perm[new_id] = old_id``:
+    train in the permuted space, then map weights back with
+    ``unpermute_weights(w, perm)``. Power-law column distributions benefit;
+    for uniformly random columns the permutation is a measured no-op by
+    symmetry (every renumbering is distribution-identical — see BACKLOG).
+    Deterministic (stable sorts throughout).
+    """
+    counts = torch.bincount(shard.col.to(torch.int64), minlength=shard.d)
+    # perm[new] = old, hottest first; stable for reproducibility
+    perm = torch.argsort(counts, descending=True, stable=True)
+    inv = torch.empty_like(perm)
+    inv[perm] = torch.arange(shard.d, device=perm.device, dtype=perm.dtype)
+    new_col = inv[shard.col.to(torch.int64)]
+    # re-sort columns within each row (keeps row-local gathers line-adjacent)
+    n = shard.n
+    counts_row = torch.diff(shard.rowptr.to(torch.int64))
+    rows = torch.repeat_interleave(
+        torch.arange(n, device=new_col.device, dtype=torch.int64), counts_row)
+    key = rows * shard.d + new_col
+    order = torch.argsort(key, stable=True)
+    out = CSRShard(shard.rowptr, new_col[order].to(torch.int32),
+                   shard.val[order], shard.labels, shard.d,
+                   deterministic=shard.csc is not None,
+                   sample_weight=shard.sample_weight)
+    return out, perm
+
+
+def unpermute_weights(w: torch.Tensor, perm: torch.Tensor) -> torch.Tensor:
+    """Map weights trained on a column-clustered shard back to the original
+    column IDs: out[perm[new]] = w[new]."""
+    out = torch.empty_like(w)
+    out[perm] = w
+    return out
+
+
+def permute_weights(w: torch.Tensor, perm: torch.Tensor) -> torch.Tensor:
+    """Original-space weights -> clustered space (e.g. warm starts)."""
+    return w[perm].contiguous()

@@ -115,3 +115,41 @@ def test_mixed_agd_gpu():
     assert len(h_g) == len(h_c)
     for a, b in zip(h_g, h_c):
         assert abs(a - b) < 1e-3 * max(1.0, abs(b)), (a, b)
+
+
+def test_reindex_columns_permutation_equivariance():
+    """Column-frequency clustering is a pure renumbering: losses identical,
+    gradients/weights permuted by exactly the returned perm, CSC rebuilt."""
+    from sparkagd_amd.data import (generate_csr_problem, permute_weights,
+                                   reindex_columns, unpermute_weights)
+
+    shard, _ = generate_csr_problem(3000, 500, 12, seed=71, col_dist="zipf",
+                                    zipf_a=1.2)
+    clustered, perm = reindex_columns(shard)
+    assert clustered.nnz == shard.nnz and clustered.d == shard.d
+    # hot columns got low IDs: counts must be non-increasing
+    counts = torch.bincount(clustered.col.to(torch.int64), minlength=500)
+    assert torch.all(counts[:-1] >= counts[1:])
+
+    torch.manual_seed(5)
+    w = torch.randn(500, dtype=torch.float32) * 0.1
+    wp = permute_weights(w, perm)
+    g0, lc0 = shard.eval(w, LogisticGradient.LOSS_TYPE)
+    g1, lc1 = clustered.eval(wp, LogisticGradient.LOSS_TYPE)
+    torch.testing.assert_close(lc0, lc1, rtol=1e-9, atol=1e-9)
+    torch.testing.assert_close(unpermute_weights(g1, perm), g0,
+                               rtol=1e-5, atol=1e-6)
+    # margins are row-space quantities: identical under the renumbering
+    torch.testing.assert_close(clustered.margins(wp), shard.margins(w),
+                               rtol=1e-5, atol=1e-6)
+
+    # full AGD solve in clustered space maps back to the original solution
+    w0 = torch.zeros(500, dtype=torch.float32)
+    w_o, h_o = run(shard, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
+                   0.05, w0, 1.0, math.inf, 0.5, 0.9, True)
+    w_c, h_c = run(clustered, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
+                   0.05, w0, 1.0, math.inf, 0.5, 0.9, True)
+    for a, b in zip(h_o, h_c):
+        assert abs(a - b) < 1e-6 * max(1.0, abs(b))
+    torch.testing.assert_close(unpermute_weights(w_c, perm), w_o,
+                               rtol=1e-4, atol=1e-6)
