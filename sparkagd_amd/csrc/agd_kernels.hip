@@ -823,6 +823,87 @@ __global__ __launch_bounds__(BLOCK) void k_csc_grad(
   }
 }
 
+// --- Skew-robust CSC gradient (round 2) -----------------------------------
+// One thread per column serializes on power-law data: a Zipf(1.1) d=1e7
+// shard puts ~9% of ALL nnz in its hottest column, and one thread grinding
+// 6M entries turned the 2.26 ms uniform step into 943 ms (measured,
+// profiles/r02_csr_skew_ab.txt). Fix: columns with nnz <= heavy_T keep the
+// thread-per-column gather (the measured-fast path on uniform data);
+// heavier columns are split into fixed S-entry tasks reduced wave-per-task,
+// then combined IN TASK ORDER per column. Fixed segmentation + shuffle
+// reductions + sequential combine = still bitwise deterministic.
+
+__global__ __launch_bounds__(BLOCK) void k_csc_grad_light(
+    const int* __restrict__ colptr, const int* __restrict__ row,
+    const float* __restrict__ val, const float* __restrict__ mult, ll d,
+    int heavy_T, float* __restrict__ grad) {
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll c = (ll)blockIdx.x * BLOCK + threadIdx.x; c < d; c += stride) {
+    const int k_lo = colptr[c], k_hi = colptr[c + 1];
+    if (k_hi - k_lo > heavy_T) continue;  // heavy path owns grad[c]
+    float acc = 0.f;
+    for (int k = k_lo; k < k_hi; ++k) acc += val[k] * mult[row[k]];
+    grad[c] = acc;
+  }
+}
+
+// Wave per task: task t covers S entries of heavy column heavy_cols[i]
+// starting at slot (t - taskptr[i]); lanes stride the window with 16-B
+// val/row loads, wave-reduce, write partial[t].
+__global__ __launch_bounds__(BLOCK) void k_csc_heavy_partial(
+    const int* __restrict__ colptr, const int* __restrict__ row,
+    const float* __restrict__ val, const float* __restrict__ mult,
+    const int* __restrict__ heavy_cols, const int* __restrict__ taskptr,
+    const int* __restrict__ task_heavy_idx, ll n_tasks, int S,
+    float* __restrict__ partial) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
+  const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  using i32x4 = __attribute__((ext_vector_type(4))) int;
+  for (ll t = wave_gid; t < n_tasks; t += n_waves) {
+    const int i = task_heavy_idx[t];
+    const int c = heavy_cols[i];
+    const int c_lo = colptr[c], c_hi = colptr[c + 1];
+    const int k_lo = c_lo + (int)(t - taskptr[i]) * S;
+    const int k_hi = min(k_lo + S, c_hi);
+    float acc = 0.f;
+    const int head_end = (k_lo + 3) & ~3;
+    const int body_end = k_hi & ~3;
+    if (head_end + 4 * WAVE <= body_end) {
+      for (int k = k_lo + lane; k < head_end; k += WAVE)
+        acc += val[k] * mult[row[k]];
+      for (int k = head_end + 4 * lane; k + 4 <= body_end; k += 4 * WAVE) {
+        const f32x4 v = *(const f32x4*)&val[k];
+        const i32x4 r = *(const i32x4*)&row[k];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc += v[j] * mult[r[j]];
+      }
+      for (int k = body_end + lane; k < k_hi; k += WAVE)
+        acc += val[k] * mult[row[k]];
+    } else {
+      for (int k = k_lo + lane; k < k_hi; k += WAVE)
+        acc += val[k] * mult[row[k]];
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) partial[t] = acc;
+  }
+}
+
+// Thread per heavy column: sum its task partials in index order
+// (deterministic; at most ceil(nnz_c / S) terms).
+__global__ __launch_bounds__(BLOCK) void k_csc_heavy_combine(
+    const int* __restrict__ heavy_cols, const int* __restrict__ taskptr,
+    const float* __restrict__ partial, ll n_heavy, float* __restrict__ grad) {
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n_heavy; i += stride) {
+    float acc = 0.f;
+    for (int t = taskptr[i]; t < taskptr[i + 1]; ++t) acc += partial[t];
+    grad[heavy_cols[i]] = acc;
+  }
+}
+
 __global__ __launch_bounds__(BLOCK) void k_csr_grad(
     const int* __restrict__ rowptr, const int* __restrict__ col,
     const float* __restrict__ val, const float* __restrict__ mult, ll n,
@@ -1250,6 +1331,39 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
     const int grid = grid_for(n, WAVES_PER_BLOCK);
     hipLaunchKernelGGL(k_csr_grad, dim3(grid), dim3(BLOCK), 0, s, rp, ci, v,
                        mult, n, (float*)grad_out);
+  }
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+// Skew-robust deterministic CSC gradient (see k_csc_grad_light above).
+// Callers compute mult first (agd_csr_eval with need_grad=0), then invoke
+// this with the heavy-column task structure built at shard construction.
+extern "C" int agd_csc_grad_skew(const void* colptr, const void* row,
+                                 const void* val, const void* mult,
+                                 long long d, int heavy_T,
+                                 const void* heavy_cols, const void* taskptr,
+                                 const void* task_heavy_idx,
+                                 long long n_heavy, long long n_tasks, int S,
+                                 void* partial, void* grad, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  {
+    const int grid = grid_for(d, BLOCK);
+    hipLaunchKernelGGL(k_csc_grad_light, dim3(grid), dim3(BLOCK), 0, s,
+                       (const int*)colptr, (const int*)row, (const float*)val,
+                       (const float*)mult, d, heavy_T, (float*)grad);
+  }
+  if (n_tasks > 0) {
+    const int grid = grid_for(n_tasks, WAVES_PER_BLOCK);
+    hipLaunchKernelGGL(k_csc_heavy_partial, dim3(grid), dim3(BLOCK), 0, s,
+                       (const int*)colptr, (const int*)row, (const float*)val,
+                       (const float*)mult, (const int*)heavy_cols,
+                       (const int*)taskptr, (const int*)task_heavy_idx,
+                       n_tasks, S, (float*)partial);
+    const int grid2 = grid_for(n_heavy, BLOCK);
+    hipLaunchKernelGGL(k_csc_heavy_combine, dim3(grid2), dim3(BLOCK), 0, s,
+                       (const int*)heavy_cols, (const int*)taskptr,
+                       (const float*)partial, n_heavy, (float*)grad);
   }
   HIP_CHECK(hipGetLastError());
   return 0;

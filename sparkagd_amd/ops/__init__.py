@@ -101,10 +101,11 @@ def csr_eval(
     csc=None,
     need_grad: bool = True,
     sample_weight: Optional[torch.Tensor] = None,
+    csc_heavy=None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _use_hip(val):
         return _get_hip().csr_eval(rowptr, col, val, labels, w, loss_type, mask,
-                                   d, csc, need_grad, sample_weight)
+                                   d, csc, need_grad, sample_weight, csc_heavy)
     # the torch reference (sparse_csr @ / .t() @) is already deterministic
     return reference.csr_eval(rowptr, col, val, labels, w, loss_type, mask, d,
                               need_grad, sample_weight)
@@ -150,11 +151,12 @@ def csr_margins(rowptr, col, val, v):
 
 def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
                           mask=None, d=None, csc=None, need_grad=True,
-                          sample_weight=None):
+                          sample_weight=None, csc_heavy=None):
     if _use_hip(val):
         return _get_hip().csr_eval_from_margins(rowptr, col, val, margins, labels,
                                                 loss_type, mask, d, csc,
-                                                need_grad, sample_weight)
+                                                need_grad, sample_weight,
+                                                csc_heavy)
     return reference.csr_eval_from_margins(rowptr, col, val, margins, labels,
                                            loss_type, mask, d, csc, need_grad,
                                            sample_weight)

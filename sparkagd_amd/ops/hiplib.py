@@ -82,6 +82,8 @@ def load() -> ctypes.CDLL:
     lib.agd_csr_margins_multi.argtypes = [P, P, P, P, LL, I, I, P, P]
     lib.agd_csc_grad_multi.restype = I
     lib.agd_csc_grad_multi.argtypes = [P, P, P, P, LL, I, P, P]
+    lib.agd_csc_grad_skew.restype = I
+    lib.agd_csc_grad_skew.argtypes = [P, P, P, P, LL, I, P, P, P, LL, LL, I, P, P, P]
 
     _lib = lib
     return lib
@@ -301,6 +303,24 @@ def dense_grad_from_mult(features: torch.Tensor, mult: torch.Tensor) -> torch.Te
     return grad
 
 
+def _csc_grad_skew(csc, csc_heavy, mult: torch.Tensor, d: int) -> torch.Tensor:
+    """Deterministic skew-robust CSC gradient: light thread-per-column pass
+    for columns <= heavy_T nnz + wave-per-task partials for split heavy
+    columns, combined in task order (agd_csc_grad_skew)."""
+    lib = load()
+    colptr, crow, cval = csc
+    grad = torch.empty(d, dtype=torch.float32, device=cval.device)
+    rc = lib.agd_csc_grad_skew(
+        _ptr(colptr), _ptr(crow), _ptr(cval), _ptr(mult.contiguous()), d,
+        int(csc_heavy["heavy_T"]), _ptr(csc_heavy["cols"]),
+        _ptr(csc_heavy["taskptr"]), _ptr(csc_heavy["task_idx"]),
+        csc_heavy["cols"].numel(), csc_heavy["task_idx"].numel(),
+        int(csc_heavy["S"]), _ptr(csc_heavy["partial"]), _ptr(grad),
+        _stream(cval))
+    _check(rc)
+    return grad
+
+
 def csr_eval(
     rowptr: torch.Tensor,
     col: torch.Tensor,
@@ -313,6 +333,7 @@ def csr_eval(
     csc: Optional[Tuple[torch.Tensor, torch.Tensor, torch.Tensor]] = None,
     need_grad: bool = True,
     sample_weight: Optional[torch.Tensor] = None,
+    csc_heavy: Optional[dict] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     lib = load()
     assert val.is_cuda and val.dtype == torch.float32
@@ -346,14 +367,19 @@ def csr_eval(
     mult = torch.empty(n, dtype=torch.float32, device=dev)
 
     sw = _prep_weights(sample_weight, val.device)
+    # skewed columns: let the C call stop after the multiplier, then run the
+    # skew-robust gradient on the mult buffer
+    skew = need_grad and csc is not None and csc_heavy is not None
     rc = lib.agd_csr_eval(
         _ptr(rowptr), _ptr(col), _ptr(val), _ptr(labels), _ptr(mask), _ptr(sw),
         _ptr(w.contiguous()), n, val.numel(), d, _ptr(grad), _ptr(loss_count),
         _ptr(margins), _ptr(mult), loss_type,
-        _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 0,
+        _ptr(cp), _ptr(cr), _ptr(cv), 1 if (need_grad and not skew) else 0, 0,
         _ptr(_red_ws(val.device)), _stream(val),
     )
     _check(rc)
+    if skew:
+        grad = _csc_grad_skew(csc, csc_heavy, mult, d)
     return grad, loss_count
 
 
@@ -557,7 +583,7 @@ def csr_margins(rowptr, col, val, v: torch.Tensor) -> torch.Tensor:
 
 def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
                           mask=None, d=None, csc=None, need_grad=True,
-                          sample_weight=None):
+                          sample_weight=None, csc_heavy=None):
     lib = load()
     n = rowptr.numel() - 1
     rowptr, col = _csr_idx(rowptr, col)
@@ -576,14 +602,18 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
     loss_count = torch.zeros(2, dtype=torch.float64, device=dev)
     mult = torch.empty(n, dtype=torch.float32, device=dev)
     sw = _prep_weights(sample_weight, val.device)
+    skew = need_grad and csc is not None and csc_heavy is not None
     rc = lib.agd_csr_eval(
         _ptr(rowptr), _ptr(col), _ptr(val.contiguous()),
         _ptr(labels), _ptr(mask), _ptr(sw), None, n, val.numel(), d,
         _ptr(grad), _ptr(loss_count), _ptr(margins.contiguous()), _ptr(mult),
-        loss_type, _ptr(cp), _ptr(cr), _ptr(cv), 1 if need_grad else 0, 2,
+        loss_type, _ptr(cp), _ptr(cr), _ptr(cv),
+        1 if (need_grad and not skew) else 0, 2,
         _ptr(_red_ws(val.device)), _stream(val),
     )
     _check(rc)
+    if skew:
+        grad = _csc_grad_skew(csc, csc_heavy, mult, d)
     return grad, loss_count
 
 

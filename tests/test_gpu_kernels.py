@@ -496,3 +496,52 @@ def test_multiplier_multi_anyk_matches_oracle(k):
         torch.testing.assert_close(lc, lc_ref, rtol=1e-6, atol=1e-6)
         Mb, lcb = hiplib.multiplier_multi(z.reshape(-1), y, k, kc, m_arg, w_arg)
         assert torch.equal(M, Mb) and torch.equal(lc, lcb)
+
+
+def test_csc_grad_skew_matches_oracle(monkeypatch):
+    """Skew-robust CSC gradient (heavy-column split + in-order combine) vs
+    the torch oracle on a deliberately skewed shard, with the thresholds
+    lowered so the heavy path engages at test scale; bitwise determinism
+    and equality-of-structure with the unsplit path."""
+    from sparkagd_amd.data import CSRShard
+    from sparkagd_amd.ops import reference as ref
+
+    monkeypatch.setattr(CSRShard, "CSC_HEAVY_T", 128)
+    monkeypatch.setattr(CSRShard, "CSC_TASK_S", 128)
+
+    g = torch.Generator(device=DEV).manual_seed(17)
+    n, d, nnz_per_row = 30000, 5000, 16
+    nnz = n * nnz_per_row
+    # ~40% of nnz land in 8 hot columns, rest uniform
+    hot = torch.randint(0, 8, (nnz,), generator=g, device=DEV, dtype=torch.int32)
+    uni = torch.randint(0, d, (nnz,), generator=g, device=DEV, dtype=torch.int32)
+    pick = torch.rand(nnz, generator=g, device=DEV) < 0.4
+    col = torch.where(pick, hot * 601 % d, uni).to(torch.int32)
+    col = col.view(n, nnz_per_row).sort(dim=1).values.reshape(-1)
+    val = torch.randn(nnz, generator=g, device=DEV)
+    rowptr = torch.arange(0, nnz + 1, nnz_per_row, device=DEV, dtype=torch.int32)
+    labels = (torch.rand(n, generator=g, device=DEV) < 0.5).float()
+    shard = CSRShard(rowptr, col, val, labels, d)
+    assert shard.csc_heavy is not None, "heavy path must engage on this shard"
+    assert shard.csc_heavy["task_idx"].numel() > shard.csc_heavy["cols"].numel()
+
+    w = torch.randn(d, generator=g, device=DEV) * 0.05
+    gh, lh = shard.eval(w, 0)  # logistic
+    gr, lr = ref.csr_eval(shard.rowptr, shard.col, shard.val, labels, w, 0,
+                          None, d)
+    torch.testing.assert_close(lh, lr.to(lh.dtype), rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(gh, gr.to(gh.dtype), rtol=2e-4, atol=2e-4)
+    gh2, lh2 = shard.eval(w, 0)
+    assert torch.equal(gh, gh2) and torch.equal(lh, lh2)
+
+    # unsplit (thresholds back to default => heavy structure absent at this
+    # scale) must agree to fp-reassociation level
+    monkeypatch.setattr(CSRShard, "CSC_HEAVY_T", 2048)
+    monkeypatch.setattr(CSRShard, "CSC_TASK_S", 2048)
+    shard2 = CSRShard(rowptr, col, val, labels, d)
+    assert shard2.csc_heavy is not None  # hottest col >> 2048 here too
+    monkeypatch.setattr(CSRShard, "CSC_HEAVY_T", 10**9)
+    shard3 = CSRShard(rowptr, col, val, labels, d)
+    assert shard3.csc_heavy is None
+    g3, l3 = shard3.eval(w, 0)
+    torch.testing.assert_close(gh, g3, rtol=1e-4, atol=1e-4)
