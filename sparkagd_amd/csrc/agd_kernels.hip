@@ -1689,15 +1689,75 @@ __global__ __launch_bounds__(BLOCK) void k_csr_margins_multi(
   }
 }
 
+// Multiclass analogs of the skew-robust split (see k_csc_grad_light):
+// wave per task with KC accumulators, partials combined in task order.
+template <int KC>
+__global__ __launch_bounds__(BLOCK) void k_csc_heavy_partial_multi(
+    const int* __restrict__ colptr, const int* __restrict__ row,
+    const float* __restrict__ val, const float* __restrict__ M,
+    const int* __restrict__ heavy_cols, const int* __restrict__ taskptr,
+    const int* __restrict__ task_heavy_idx, ll n_tasks, int S,
+    float* __restrict__ partial) {
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
+  const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
+  for (ll t = wave_gid; t < n_tasks; t += n_waves) {
+    const int i = task_heavy_idx[t];
+    const int c = heavy_cols[i];
+    const int k_lo = colptr[c] + (int)(t - taskptr[i]) * S;
+    const int k_hi = min(k_lo + S, colptr[c + 1]);
+    float acc[KC];
+#pragma unroll
+    for (int j = 0; j < KC; ++j) acc[j] = 0.f;
+    for (int k = k_lo + lane; k < k_hi; k += WAVE) {
+      const float v = val[k];
+      const float* __restrict__ mr = M + (ll)row[k] * KC;
+#pragma unroll
+      for (int ch = 0; ch < KC / 4; ++ch) {
+        const f32x4 mv = *(const f32x4*)(mr + ch * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[ch * 4 + j] += v * mv[j];
+      }
+    }
+    float* __restrict__ dst = partial + t * KC;
+#pragma unroll
+    for (int j = 0; j < KC; ++j) {
+      const float s = wave_reduce_sum(acc[j]);
+      if (lane == 0) dst[j] = s;
+    }
+  }
+}
+
+template <int KC>
+__global__ __launch_bounds__(BLOCK) void k_csc_heavy_combine_multi(
+    const int* __restrict__ heavy_cols, const int* __restrict__ taskptr,
+    const float* __restrict__ partial, ll n_heavy, float* __restrict__ grad) {
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n_heavy; i += stride) {
+    float acc[KC];
+#pragma unroll
+    for (int j = 0; j < KC; ++j) acc[j] = 0.f;
+    for (int t = taskptr[i]; t < taskptr[i + 1]; ++t)
+#pragma unroll
+      for (int j = 0; j < KC; ++j) acc[j] += partial[(ll)t * KC + j];
+    float* __restrict__ gr = grad + (ll)heavy_cols[i] * KC;
+#pragma unroll
+    for (int j = 0; j < KC; ++j) gr[j] = acc[j];
+  }
+}
+
 template <int KC>
 __global__ __launch_bounds__(BLOCK) void k_csc_grad_multi(
     const int* __restrict__ colptr, const int* __restrict__ row,
     const float* __restrict__ val, const float* __restrict__ M, ll d,
-    float* __restrict__ grad) {
+    int heavy_T, float* __restrict__ grad) {
   using f32x4 = __attribute__((ext_vector_type(4))) float;
   const ll stride = (ll)gridDim.x * BLOCK;
   for (ll c = (ll)blockIdx.x * BLOCK + threadIdx.x; c < d; c += stride) {
     const int k_lo = colptr[c], k_hi = colptr[c + 1];
+    if (k_hi - k_lo > heavy_T) continue;  // heavy path owns grad[c]
     float acc[KC];
 #pragma unroll
     for (int j = 0; j < KC; ++j) acc[j] = 0.f;
@@ -1759,15 +1819,38 @@ extern "C" int agd_csr_margins_multi(const void* rowptr, const void* col,
   return 0;
 }
 
+// Heavy-column args mirror agd_csc_grad_skew; n_tasks == 0 (or null heavy
+// pointers) keeps the plain thread-per-column path for every column.
 extern "C" int agd_csc_grad_multi(const void* colptr, const void* row,
                                   const void* val, const void* M, long long d,
-                                  int kc, void* grad, void* stream) {
+                                  int kc, void* grad, int heavy_T,
+                                  const void* heavy_cols, const void* taskptr,
+                                  const void* task_heavy_idx,
+                                  long long n_heavy, long long n_tasks, int S,
+                                  void* partial, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   const int grid = grid_for(d, BLOCK);
-#define LAUNCH_CG(KCV)                                                        \
-  hipLaunchKernelGGL((k_csc_grad_multi<KCV>), dim3(grid), dim3(BLOCK), 0, s,  \
-                     (const int*)colptr, (const int*)row, (const float*)val,  \
-                     (const float*)M, d, (float*)grad)
+  const int light_T = (n_tasks > 0) ? heavy_T : 0x7fffffff;
+#define LAUNCH_CG(KCV)                                                         \
+  do {                                                                         \
+    hipLaunchKernelGGL((k_csc_grad_multi<KCV>), dim3(grid), dim3(BLOCK), 0, s, \
+                       (const int*)colptr, (const int*)row, (const float*)val, \
+                       (const float*)M, d, light_T, (float*)grad);             \
+    if (n_tasks > 0) {                                                         \
+      const int g1 = grid_for(n_tasks, WAVES_PER_BLOCK);                       \
+      hipLaunchKernelGGL((k_csc_heavy_partial_multi<KCV>), dim3(g1),           \
+                         dim3(BLOCK), 0, s, (const int*)colptr,                \
+                         (const int*)row, (const float*)val, (const float*)M,  \
+                         (const int*)heavy_cols, (const int*)taskptr,          \
+                         (const int*)task_heavy_idx, n_tasks, S,               \
+                         (float*)partial);                                     \
+      const int g2 = grid_for(n_heavy, BLOCK);                                 \
+      hipLaunchKernelGGL((k_csc_heavy_combine_multi<KCV>), dim3(g2),           \
+                         dim3(BLOCK), 0, s, (const int*)heavy_cols,            \
+                         (const int*)taskptr, (const float*)partial, n_heavy,  \
+                         (float*)grad);                                        \
+    }                                                                          \
+  } while (0)
   switch (kc) {
     case 4: LAUNCH_CG(4); break;
     case 8: LAUNCH_CG(8); break;

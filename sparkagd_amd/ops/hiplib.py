@@ -81,7 +81,8 @@ def load() -> ctypes.CDLL:
     lib.agd_csr_margins_multi.restype = I
     lib.agd_csr_margins_multi.argtypes = [P, P, P, P, LL, I, I, P, P]
     lib.agd_csc_grad_multi.restype = I
-    lib.agd_csc_grad_multi.argtypes = [P, P, P, P, LL, I, P, P]
+    lib.agd_csc_grad_multi.argtypes = [P, P, P, P, LL, I, P,
+                                       I, P, P, P, LL, LL, I, P, P]
     lib.agd_csc_grad_skew.restype = I
     lib.agd_csc_grad_skew.argtypes = [P, P, P, P, LL, I, P, P, P, LL, LL, I, P, P, P]
 
@@ -504,15 +505,27 @@ def csr_margins_multi(rowptr, col, val, wflat: torch.Tensor, k: int,
 
 
 def csc_grad_multi(colptr, row, cval, M: torch.Tensor, d: int,
-                   kc: int) -> torch.Tensor:
-    """grad flat [d*KC] = A^T·M via the deterministic CSC gather."""
+                   kc: int, csc_heavy: Optional[dict] = None) -> torch.Tensor:
+    """grad flat [d*KC] = A^T·M via the deterministic CSC gather; skewed
+    columns (csc_heavy, built by CSRShard) run the wave-per-task split."""
     lib = load()
     grad = torch.empty(d * kc, dtype=torch.float32, device=cval.device)
-    rc = lib.agd_csc_grad_multi(_ptr(colptr.contiguous()),
-                                _ptr(row.contiguous()),
-                                _ptr(cval.contiguous()),
-                                _ptr(M.contiguous()), d, kc, _ptr(grad),
-                                _stream(cval))
+    if csc_heavy is not None:
+        n_tasks = csc_heavy["task_idx"].numel()
+        partial = torch.empty(n_tasks * kc, dtype=torch.float32,
+                              device=cval.device)
+        rc = lib.agd_csc_grad_multi(
+            _ptr(colptr.contiguous()), _ptr(row.contiguous()),
+            _ptr(cval.contiguous()), _ptr(M.contiguous()), d, kc, _ptr(grad),
+            int(csc_heavy["heavy_T"]), _ptr(csc_heavy["cols"]),
+            _ptr(csc_heavy["taskptr"]), _ptr(csc_heavy["task_idx"]),
+            csc_heavy["cols"].numel(), n_tasks, int(csc_heavy["S"]),
+            _ptr(partial), _stream(cval))
+    else:
+        rc = lib.agd_csc_grad_multi(
+            _ptr(colptr.contiguous()), _ptr(row.contiguous()),
+            _ptr(cval.contiguous()), _ptr(M.contiguous()), d, kc, _ptr(grad),
+            0, None, None, None, 0, 0, 0, None, _stream(cval))
     _check(rc)
     return grad
 

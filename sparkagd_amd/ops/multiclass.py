@@ -308,7 +308,8 @@ def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
                 "(CSRShard(..., deterministic=True))")
         colptr, crow, cval = shard.csc
         if kc <= 32:
-            gradp = hiplib.csc_grad_multi(colptr, crow, cval, M, shard.d, kc)
+            gradp = hiplib.csc_grad_multi(colptr, crow, cval, M, shard.d, kc,
+                                          getattr(shard, 'csc_heavy', None))
             if kc != k:
                 return gradp.reshape(shard.d, kc)[:, :k].reshape(-1).contiguous(), lc
             return gradp, lc
@@ -322,7 +323,8 @@ def eval_multi_csr_from_margins(shard, margins_padded_flat, k,
                                    device=shard.val.device)
             mc_chunk[:, : hi - lo] = m2d[:, lo:hi]
             gc = hiplib.csc_grad_multi(colptr, crow, cval,
-                                       mc_chunk.reshape(-1), shard.d, kcc)
+                                       mc_chunk.reshape(-1), shard.d, kcc,
+                                       getattr(shard, 'csc_heavy', None))
             grad[:, lo:hi] = gc.reshape(shard.d, kcc)[:, : hi - lo]
         return grad.reshape(-1).contiguous(), lc
     z = margins_padded_flat.reshape(n, kc)[:, :k]

@@ -545,3 +545,41 @@ def test_csc_grad_skew_matches_oracle(monkeypatch):
     assert shard3.csc_heavy is None
     g3, l3 = shard3.eval(w, 0)
     torch.testing.assert_close(gh, g3, rtol=1e-4, atol=1e-4)
+
+
+def test_csc_grad_multi_skew_matches_oracle(monkeypatch):
+    """Multiclass skew-robust CSC gradient vs the torch oracle on a skewed
+    shard (k_csc_heavy_partial_multi / k_csc_heavy_combine_multi)."""
+    from sparkagd_amd.data import CSRShard
+    from sparkagd_amd.ops import multiclass as mc
+
+    monkeypatch.setattr(CSRShard, "CSC_HEAVY_T", 128)
+    monkeypatch.setattr(CSRShard, "CSC_TASK_S", 128)
+    K = 6
+    g = torch.Generator(device=DEV).manual_seed(19)
+    n, d, nnz_per_row = 20000, 4000, 12
+    nnz = n * nnz_per_row
+    hot = torch.randint(0, 5, (nnz,), generator=g, device=DEV, dtype=torch.int32)
+    uni = torch.randint(0, d, (nnz,), generator=g, device=DEV, dtype=torch.int32)
+    pick = torch.rand(nnz, generator=g, device=DEV) < 0.35
+    col = torch.where(pick, hot * 797 % d, uni).to(torch.int32)
+    col = col.view(n, nnz_per_row).sort(dim=1).values.reshape(-1)
+    val = torch.randn(nnz, generator=g, device=DEV)
+    rowptr = torch.arange(0, nnz + 1, nnz_per_row, device=DEV, dtype=torch.int32)
+    labels = torch.randint(0, K, (n,), generator=g, device=DEV).float()
+    shard = CSRShard(rowptr, col, val, labels, d)
+    assert shard.csc_heavy is not None
+
+    from sparkagd_amd import MultinomialLogisticGradient
+
+    W = (torch.randn(d * K, generator=g, device=DEV) / 4.0).contiguous()
+    grad = MultinomialLogisticGradient(K)
+    gh, lh = grad.eval(shard, W)
+    zf = mc.ref_csr_margins_multi(shard.rowptr, shard.col, shard.val, W, K,
+                                  d).reshape(-1, K)
+    m, lr = mc.ref_multiplier_multi(zf, labels)
+    gr = mc.ref_csr_grad_multi(shard.rowptr, shard.col, shard.val, m, d)
+    torch.testing.assert_close(lh, lr, rtol=1e-6, atol=1e-6)
+    torch.testing.assert_close(gh, gr, rtol=3e-4, atol=3e-4)
+    gh2, lh2 = grad.eval(shard, W)
+    assert torch.equal(gh, gh2)
