@@ -59,13 +59,33 @@ class GramOperator:
 
     def __init__(self, shard: DenseShard, comm: Communicator,
                  chunk_rows: int = 1024, d_chunk: int = 65536,
-                 mem_budget_bytes: int = 64 << 30):
+                 mem_budget_bytes: int = 64 << 30,
+                 k_dtype: Optional[str] = None):
         t0 = time.perf_counter()
+        if k_dtype is None:  # env override for A/B runs (bench.py)
+            import os
+
+            k_dtype = os.environ.get("SPARKAGD_GRAM_K", "auto")
         self.comm = comm
         A = shard.features
         dev = A.device
         acc = torch.float64 if A.dtype == torch.float64 else torch.float32
         self.acc = acc
+        # K storage dtype. The per-trial K·m GEMV streams the whole K at the
+        # memory roofline (n_local*n_global*itemsize bytes), so storing K in
+        # bf16 halves trial time — and matches the direct path's dtype
+        # discipline exactly (the direct solver streams bf16 A with f32
+        # accumulation; gm from a bf16 K carries the same class of rounding
+        # as margins from bf16 features). The coefficient-space algebra
+        # stays self-consistent: G is built from f64 dots of the margins
+        # actually stored, and the final x materializes from exact
+        # A^T(sum c_j m_j)/c. 'auto' = bf16 iff the shard is bf16 (GPU);
+        # 'f32' forces full precision (and is the rule off-GPU/f64).
+        if k_dtype not in ("auto", "bf16", "f32"):
+            raise ValueError("k_dtype must be auto|bf16|f32")
+        use_bf16_k = (A.dtype == torch.bfloat16 and A.is_cuda
+                      and k_dtype in ("auto", "bf16"))
+        self.k_dtype = torch.bfloat16 if use_bf16_k else acc
         n_local, d = A.shape
         counts = [n_local]
         if comm.world_size > 1:
@@ -79,7 +99,7 @@ class GramOperator:
         n_global = int(self.offsets[-1])
         self.n_local, self.n_global = n_local, n_global
 
-        k_bytes = n_local * n_global * acc.itemsize
+        k_bytes = n_local * n_global * self.k_dtype.itemsize
         if k_bytes > mem_budget_bytes:
             raise MemoryError(
                 f"Gram matrix needs {k_bytes/2**30:.1f} GiB > budget "
@@ -89,8 +109,8 @@ class GramOperator:
         # (fp32 accumulation; ~15x the chunked f32 rocBLAS route, no cast
         # traffic). Other dtypes/devices: chunked f32/f64 torch GEMMs.
         use_lt = A.dtype == torch.bfloat16 and A.is_cuda
-        K = torch.empty((n_local, n_global), dtype=acc, device=dev)
-        if use_lt and comm.world_size == 1:
+        K = torch.empty((n_local, n_global), dtype=self.k_dtype, device=dev)
+        if use_lt and comm.world_size == 1 and self.k_dtype != torch.bfloat16:
             from .ops.hiplib import gemm_bf16f32_nt
 
             gemm_bf16f32_nt(A, A, K)  # one bf16->f32 GEMM, C written in place
@@ -116,24 +136,35 @@ class GramOperator:
                     block = torch.empty((n_local, c_hi - c_lo), dtype=torch.float32,
                                         device=dev)
                     gemm_bf16f32_nt(A, buf.contiguous(), block)
-                    dst.copy_(block)
+                    dst.copy_(block)  # f32 accumulation; rounded here iff K is bf16
                     del block
                 else:
-                    dst.zero_()
+                    tmp = torch.zeros((n_local, c_hi - c_lo), dtype=acc, device=dev)
                     for d_lo in range(0, d, d_chunk):
                         d_hi = min(d_lo + d_chunk, d)
-                        dst.addmm_(A[:, d_lo:d_hi].to(acc), buf[:, d_lo:d_hi].to(acc).T)
+                        tmp.addmm_(A[:, d_lo:d_hi].to(acc), buf[:, d_lo:d_hi].to(acc).T)
+                    dst.copy_(tmp)
+                    del tmp
                 del buf
         self.K = K.contiguous()
         self.build_seconds = time.perf_counter() - t0
 
     def matvec(self, m_global: torch.Tensor, ncols: int = 1) -> torch.Tensor:
         """K @ m_global -> local margin slice (flat). ncols > 1 (multiclass:
-        padded class columns) runs one rocBLAS GEMM instead of the GEMV
-        margins kernel; zero pad columns stay exactly zero."""
+        padded class columns) runs one GEMM instead of the GEMV margins
+        kernel; zero pad columns stay exactly zero (bf16 rounding of an
+        exact zero is an exact zero)."""
         if ncols > 1:
             mg = m_global.to(self.acc).reshape(self.n_global, ncols)
+            if self.K.dtype == torch.bfloat16:
+                from .ops.hiplib import gemm_bf16f32_nt
+
+                z = torch.empty((self.n_local, ncols), dtype=torch.float32,
+                                device=self.K.device)
+                gemm_bf16f32_nt(self.K, mg.T.to(torch.bfloat16).contiguous(), z)
+                return z.reshape(-1)
             return (self.K @ mg).reshape(-1)
+        # the margins kernel streams bf16/f32/f64 K with f32/f64 m
         return ops.dense_margins(self.K, m_global.to(self.acc))
 
     def all_gather_m(self, m_local: torch.Tensor, ncols: int = 1) -> torch.Tensor:
