@@ -438,6 +438,39 @@ __device__ __forceinline__ float softplus<float>(float t) {
   return log1pf(__expf(t));
 }
 
+// Per-example multiplier + loss for the binary losses (shared by the
+// margins-array and affine-combination kernels below).
+template <typename TACC>
+__device__ __forceinline__ void loss_mult(int loss_type, TACC z, TACC y,
+                                          TACC& m, TACC& l) {
+  if (loss_type == LOSS_LOGISTIC) {
+    m = (TACC)1 / ((TACC)1 + exp(-z)) - y;
+    l = (y > (TACC)0) ? softplus<TACC>(-z) : softplus<TACC>(z);
+  } else if (loss_type == LOSS_LSQ) {
+    const TACC diff = z - y;
+    m = (TACC)2 * diff;
+    l = diff * diff;
+  } else if (loss_type == LOSS_HINGE) {
+    const TACC s = (TACC)2 * y - (TACC)1;
+    const TACC sz = s * z;
+    m = (sz < (TACC)1) ? -s : (TACC)0;
+    l = (sz < (TACC)1) ? (TACC)1 - sz : (TACC)0;
+  } else {  // LOSS_SMOOTH_HINGE: 0 if sz>=1; (1-sz)^2/2 if 0<sz<1; 0.5-sz else
+    const TACC s = (TACC)2 * y - (TACC)1;
+    const TACC sz = s * z;
+    if (sz >= (TACC)1) {
+      m = (TACC)0;
+      l = (TACC)0;
+    } else if (sz > (TACC)0) {
+      m = -s * ((TACC)1 - sz);
+      l = (TACC)0.5 * ((TACC)1 - sz) * ((TACC)1 - sz);
+    } else {
+      m = -s;
+      l = (TACC)0.5 - sz;
+    }
+  }
+}
+
 template <typename TACC>
 __global__ __launch_bounds__(BLOCK) void k_multiplier(
     const TACC* __restrict__ margins, const float* __restrict__ labels,
@@ -455,32 +488,7 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
     for (int s = 1; s < n_slabs; ++s) z += margins[(ll)s * n + i];
     const TACC y = (TACC)labels[i];
     TACC m, l;
-    if (loss_type == LOSS_LOGISTIC) {
-      m = (TACC)1 / ((TACC)1 + exp(-z)) - y;
-      l = (y > (TACC)0) ? softplus<TACC>(-z) : softplus<TACC>(z);
-    } else if (loss_type == LOSS_LSQ) {
-      const TACC diff = z - y;
-      m = (TACC)2 * diff;
-      l = diff * diff;
-    } else if (loss_type == LOSS_HINGE) {
-      const TACC s = (TACC)2 * y - (TACC)1;
-      const TACC sz = s * z;
-      m = (sz < (TACC)1) ? -s : (TACC)0;
-      l = (sz < (TACC)1) ? (TACC)1 - sz : (TACC)0;
-    } else {  // LOSS_SMOOTH_HINGE: 0 if sz>=1; (1-sz)^2/2 if 0<sz<1; 0.5-sz else
-      const TACC s = (TACC)2 * y - (TACC)1;
-      const TACC sz = s * z;
-      if (sz >= (TACC)1) {
-        m = (TACC)0;
-        l = (TACC)0;
-      } else if (sz > (TACC)0) {
-        m = -s * ((TACC)1 - sz);
-        l = (TACC)0.5 * ((TACC)1 - sz) * ((TACC)1 - sz);
-      } else {
-        m = -s;
-        l = (TACC)0.5 - sz;
-      }
-    }
+    loss_mult<TACC>(loss_type, z, y, m, l);
     if (sample_weight) {
       const TACC sw = (TACC)sample_weight[i];
       mult[i] = m * sw;
@@ -494,6 +502,71 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier(
   }
   double acc[2] = {lsum, cnt};
   block_reduce_partial<2>(acc, red_part);
+}
+
+// ---------------------------------------------------------------------------
+// Gram-solver fused trial kernels (round 2): the per-trial cost after the
+// bf16 K·m stream is ~12 small n-space launches + host algebra; these two
+// kernels collapse them. Binary losses, full batch (no masks), affine prox.
+// ---------------------------------------------------------------------------
+
+// m_y/loss at the AT interpolation y WITHOUT materializing ym:
+// z_i = a*xm[i] + b*zm[i] (a = 1-theta, b = theta).
+__global__ __launch_bounds__(BLOCK) void k_multiplier_affine(
+    const float* __restrict__ xm, const float* __restrict__ zm, double a,
+    double b, const float* __restrict__ labels,
+    const float* __restrict__ sample_weight, int loss_type, ll n,
+    float* __restrict__ mult, double* __restrict__ red_part) {
+  double lsum = 0.0, cnt = 0.0;
+  const float fa = (float)a, fb = (float)b;
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
+    const float z = fa * xm[i] + fb * zm[i];
+    const float y = labels[i];
+    float m, l;
+    loss_mult<float>(loss_type, z, y, m, l);
+    if (sample_weight) {
+      const float sw = sample_weight[i];
+      mult[i] = m * sw;
+      lsum += (double)l * (double)sw;
+      cnt += (double)sw;
+    } else {
+      mult[i] = m;
+      lsum += (double)l;
+      cnt += 1.0;
+    }
+  }
+  double acc[2] = {lsum, cnt};
+  block_reduce_partial<2>(acc, red_part);
+}
+
+// One pass over n finishing a basis registration + the AT margin updates:
+//   g        = gm_raw * inv_c          (gm_raw = K·m_global, unscaled)
+//   XB_t     = (f64) g                 (basis-margin row for the G dgemv)
+//   md       = (f64) m_y               (dgemv right-hand side)
+//   Mstore_T = m_y
+//   zm_new   = pz*zm_old + pg*g        (affine prox on margins)
+//   xm_new   = (1-theta)*xm_old + theta*zm_new
+__global__ __launch_bounds__(BLOCK) void k_gram_state_update(
+    const float* __restrict__ gm_raw, const float* __restrict__ m_y,
+    const float* __restrict__ xm_old, const float* __restrict__ zm_old,
+    double inv_c, double theta, double pz, double pg, ll n,
+    double* __restrict__ xb_t, double* __restrict__ md,
+    float* __restrict__ mstore_t, float* __restrict__ zm_new,
+    float* __restrict__ xm_new) {
+  const float fc = (float)inv_c, ft = (float)theta, fz = (float)pz,
+              fg = (float)pg;
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
+    const float g = gm_raw[i] * fc;
+    xb_t[i] = (double)g;
+    const float m = m_y[i];
+    md[i] = (double)m;
+    mstore_t[i] = m;
+    const float z = fz * zm_old[i] + fg * g;
+    zm_new[i] = z;
+    xm_new[i] = (1.0f - ft) * xm_old[i] + ft * z;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1332,6 +1405,40 @@ extern "C" int agd_csr_eval(const void* rowptr, const void* col, const void* val
     hipLaunchKernelGGL(k_csr_grad, dim3(grid), dim3(BLOCK), 0, s, rp, ci, v,
                        mult, n, (float*)grad_out);
   }
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int agd_gram_mult_affine(const void* xm, const void* zm, double a,
+                                    double b, const void* labels,
+                                    const void* sample_weight, int loss_type,
+                                    long long n, void* mult, void* loss_count,
+                                    void* red_ws, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = grid_for(n, BLOCK);
+  hipLaunchKernelGGL(k_multiplier_affine, dim3(grid), dim3(BLOCK), 0, s,
+                     (const float*)xm, (const float*)zm, a, b,
+                     (const float*)labels, (const float*)sample_weight,
+                     loss_type, n, (float*)mult, (double*)red_ws);
+  hipLaunchKernelGGL((k_reduce_partials<2>), dim3(1), dim3(BLOCK), 0, s,
+                     (double*)red_ws, grid, (double*)loss_count);
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int agd_gram_state_update(const void* gm_raw, const void* m_y,
+                                     const void* xm_old, const void* zm_old,
+                                     double inv_c, double theta, double pz,
+                                     double pg, long long n, void* xb_t,
+                                     void* md, void* mstore_t, void* zm_new,
+                                     void* xm_new, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = grid_for(n, BLOCK);
+  hipLaunchKernelGGL(k_gram_state_update, dim3(grid), dim3(BLOCK), 0, s,
+                     (const float*)gm_raw, (const float*)m_y,
+                     (const float*)xm_old, (const float*)zm_old, inv_c, theta,
+                     pz, pg, n, (double*)xb_t, (double*)md, (float*)mstore_t,
+                     (float*)zm_new, (float*)xm_new);
   HIP_CHECK(hipGetLastError());
   return 0;
 }

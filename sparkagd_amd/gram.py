@@ -313,6 +313,61 @@ def run_gram(
     backtrack_simple = True
     loss_history: List[float] = []
 
+    # Fused trial fast path (GPU, binary loss, affine prox, f32 margins):
+    # k_multiplier_affine evaluates y without materializing ym, and
+    # k_gram_state_update finishes the basis registration + both margin
+    # updates in ONE pass — ~12 small launches collapse to ~5 around the
+    # dominant K·m stream.
+    use_fused = (
+        ncols == 1
+        and dev.type == "cuda"
+        and acc == torch.float32
+        and getattr(gradient, "LOSS_TYPE", -1) >= 0
+        and updater.PROX_KIND in (ops.PROX_SIMPLE, ops.PROX_SQUARED_L2)
+        and ops._use_hip(data.features)
+    )
+    if use_fused:
+        from .ops import hiplib as _hl
+
+        md_buf = torch.empty(flat_n, dtype=torch.float64, device=dev)
+        lc_x_buf = torch.empty(2, dtype=torch.float64, device=dev)
+        sw = getattr(data, "sample_weight", None)
+        labels_f32 = data.labels.to(torch.float32).contiguous()
+
+    def fused_y_trial(th: float, L_now: float, xm_o, zm_o, cz_o, cx_o):
+        """One fused backtracking trial at theta=th: returns
+        (t_y, step, cz_t, cx_t, zm_new, xm_new, lc_y, row_dev)."""
+        nonlocal T
+        if T >= max_basis:
+            raise RuntimeError("gram basis overflow — raise max_basis")
+        m_y = torch.empty(flat_n, dtype=acc, device=dev)
+        lc_y = torch.empty(2, dtype=torch.float64, device=dev)
+        _hl.gram_mult_affine(xm_o, zm_o, 1.0 - th, th, labels_f32,
+                             gradient.LOSS_TYPE, sw, m_y, lc_y)
+        comm.allreduce_(lc_y)
+        m_global = op.all_gather_m(m_y)
+        gm_raw = op.matvec(m_global)  # unscaled K·m; state kernel applies 1/c
+        t = T + 1
+        step = 1.0 / (th * L_now)
+        pz = (1.0 - step * reg_param
+              if updater.PROX_KIND == ops.PROX_SQUARED_L2 else 1.0)
+        zm_new = torch.empty_like(zm_o)
+        xm_new = torch.empty_like(xm_o)
+        _hl.gram_state_update(gm_raw, m_y, xm_o, zm_o, 1.0 / c, th, pz, -step,
+                              XB[t], md_buf, Mstore[T], zm_new, xm_new)
+        row = XB[: t + 1] @ md_buf
+        comm.allreduce_(row)
+        T += 1
+        cz_t = prox_coeff(cz_o, t, step)
+        cx_t = (1.0 - th) * cx_o + th * cz_t
+        if backtrack_simple and beta < 1.0:
+            # loss-only x-eval through the same kernel (a=1, b=0); m_y's
+            # buffer is free again (already persisted into Mstore/md)
+            _hl.gram_mult_affine(xm_new, xm_new, 1.0, 0.0, labels_f32,
+                                 gradient.LOSS_TYPE, sw, m_y, lc_x_buf)
+            comm.allreduce_(lc_x_buf)
+        return t, step, cz_t, cx_t, zm_new, xm_new, lc_y, row
+
     def prox_coeff(cz_old: np.ndarray, t_idx: int, step: float) -> np.ndarray:
         out = cz_old.copy()
         if updater.PROX_KIND == ops.PROX_SQUARED_L2:
@@ -343,18 +398,25 @@ def run_gram(
         while True:
             theta = 2.0 / (1.0 + math.sqrt(1.0 + 4.0 * (L / L_old) / (theta_old * theta_old)))
             cy = (1.0 - theta) * cx_old + theta * cz_old
-            ym = ops.axpby(1.0 - theta, xm_old, theta, zm_old)
-            # Enqueue the whole trial's device work before the first host
-            # read: y-loss, basis registration (K·m + G row), the margin
-            # updates, and (simple mode) the x-loss — then pay ONE pipeline
-            # wait instead of three.
-            lc_y, m_y = eval_loss_async(ym)
-            t_y, gm_y, row_y = new_basis_async(m_y)
-            step = 1.0 / (theta * L)
-            cz = prox_coeff(cz_old, t_y, step)
-            cx = (1.0 - theta) * cx_old + theta * cz
-            zm = updater.prox_margins(zm_old, gm_y, step, reg_param)
-            xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
+            lc_x = None
+            if use_fused:
+                (t_y, step, cz, cx, zm, xm, lc_y, row_y) = fused_y_trial(
+                    theta, L, xm_old, zm_old, cz_old, cx_old)
+                if backtrack_simple and beta < 1.0:
+                    lc_x = lc_x_buf
+            else:
+                ym = ops.axpby(1.0 - theta, xm_old, theta, zm_old)
+                # Enqueue the whole trial's device work before the first host
+                # read: y-loss, basis registration (K·m + G row), the margin
+                # updates, and (simple mode) the x-loss — then pay ONE
+                # pipeline wait instead of three.
+                lc_y, m_y = eval_loss_async(ym)
+                t_y, gm_y, row_y = new_basis_async(m_y)
+                step = 1.0 / (theta * L)
+                cz = prox_coeff(cz_old, t_y, step)
+                cx = (1.0 - theta) * cx_old + theta * cz
+                zm = updater.prox_margins(zm_old, gm_y, step, reg_param)
+                xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
 
             if beta >= 1.0:
                 packed = torch.cat([lc_y, row_y]).cpu().numpy()
@@ -362,8 +424,7 @@ def run_gram(
                 finish_basis_host(t_y, packed[2:] / c)
                 break
 
-            lc_x = None
-            if backtrack_simple:
+            if backtrack_simple and lc_x is None:
                 lc_x, _ = eval_loss_async(xm)
             # ONE packed D2H transfer for everything this trial must read on
             # the host (y-loss, the new G row, and the pre-enqueued x-loss)

@@ -85,6 +85,10 @@ def load() -> ctypes.CDLL:
                                        I, P, P, P, LL, LL, I, P, P]
     lib.agd_csc_grad_skew.restype = I
     lib.agd_csc_grad_skew.argtypes = [P, P, P, P, LL, I, P, P, P, LL, LL, I, P, P, P]
+    lib.agd_gram_mult_affine.restype = I
+    lib.agd_gram_mult_affine.argtypes = [P, P, D, D, P, P, I, LL, P, P, P, P]
+    lib.agd_gram_state_update.restype = I
+    lib.agd_gram_state_update.argtypes = [P, P, P, P, D, D, D, D, LL, P, P, P, P, P, P]
 
     _lib = lib
     return lib
@@ -675,3 +679,49 @@ def dot_diff(x: torch.Tensor, y: torch.Tensor, g_x: torch.Tensor, g_y: torch.Ten
                           _ptr(_red_ws(x.device)), _stream(x))
     _check(rc)
     return out
+
+
+def gram_mult_affine(xm: torch.Tensor, zm: torch.Tensor, a: float, b: float,
+                     labels: torch.Tensor, loss_type: int,
+                     sample_weight: Optional[torch.Tensor] = None,
+                     mult_out: Optional[torch.Tensor] = None,
+                     lc_out: Optional[torch.Tensor] = None):
+    """(mult, loss_count) at margins a*xm + b*zm without materializing the
+    combination — the Gram solver's fused y-evaluation (binary losses,
+    full batch)."""
+    lib = load()
+    n = xm.numel()
+    dev = xm.device
+    labels = labels.contiguous()
+    if labels.dtype != torch.float32:
+        labels = labels.to(torch.float32)
+    mult = mult_out if mult_out is not None else torch.empty(
+        n, dtype=torch.float32, device=dev)
+    lc = lc_out if lc_out is not None else torch.empty(
+        2, dtype=torch.float64, device=dev)
+    lc.zero_()
+    sw = _prep_weights(sample_weight, dev)
+    rc = lib.agd_gram_mult_affine(_ptr(xm.contiguous()), _ptr(zm.contiguous()),
+                                  float(a), float(b), _ptr(labels), _ptr(sw),
+                                  loss_type, n, _ptr(mult), _ptr(lc),
+                                  _ptr(_red_ws(dev)), _stream(xm))
+    _check(rc)
+    return mult, lc
+
+
+def gram_state_update(gm_raw: torch.Tensor, m_y: torch.Tensor,
+                      xm_old: torch.Tensor, zm_old: torch.Tensor,
+                      inv_c: float, theta: float, pz: float, pg: float,
+                      xb_t: torch.Tensor, md: torch.Tensor,
+                      mstore_t: torch.Tensor, zm_new: torch.Tensor,
+                      xm_new: torch.Tensor) -> None:
+    """One fused pass finishing a Gram basis registration + the AT margin
+    updates (see k_gram_state_update)."""
+    lib = load()
+    rc = lib.agd_gram_state_update(
+        _ptr(gm_raw.contiguous()), _ptr(m_y.contiguous()),
+        _ptr(xm_old.contiguous()), _ptr(zm_old.contiguous()),
+        float(inv_c), float(theta), float(pz), float(pg), gm_raw.numel(),
+        _ptr(xb_t), _ptr(md), _ptr(mstore_t), _ptr(zm_new), _ptr(xm_new),
+        _stream(gm_raw))
+    _check(rc)
