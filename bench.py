@@ -88,6 +88,11 @@ class CountingGradient:
         self.n_evals += 1  # a loss evaluation over every example (n-space)
         return self.inner.multiplier_loss(shard, margins, mask)
 
+    def count_eval(self, n: int = 1):
+        """Called by fused solver paths that bypass the wrapped methods
+        (gram.py fused trials): n loss evaluations, zero data passes."""
+        self.n_evals += n
+
     @property
     def LOSS_TYPE(self):
         return self.inner.LOSS_TYPE
