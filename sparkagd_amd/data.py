@@ -163,9 +163,12 @@ class CSRShard:
         torch.cumsum(colcounts, dim=0, out=colptr[1:])
         return colptr.to(torch.int32).contiguous(), csc_row, csc_val
 
-    #: heavy-column split parameters (see csc_heavy above)
-    CSC_HEAVY_T = 2048
-    CSC_TASK_S = 2048
+    #: heavy-column split parameters (see csc_heavy above). The light
+    #: kernel's wave executes the MAX of its 64 threads' column lengths, so
+    #: a lower threshold moves imbalance into the wave-parallel heavy path;
+    #: env overrides allow threshold A/Bs (profiles/r02_csr_skew_ab.txt).
+    CSC_HEAVY_T = int(__import__("os").environ.get("SPARKAGD_CSC_HEAVY_T", "512"))
+    CSC_TASK_S = int(__import__("os").environ.get("SPARKAGD_CSC_TASK_S", "512"))
 
     def _build_csc_heavy(self):
         colptr = self.csc[0].to(torch.int64)
