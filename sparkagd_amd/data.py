@@ -173,9 +173,23 @@ class CSRShard:
     def _build_csc_heavy(self):
         colptr = self.csc[0].to(torch.int64)
         counts = torch.diff(colptr)
+        dev = self.val.device
+        # length-sorted visit order for the light kernel: a wave executes
+        # the MAX of its 64 threads' column lengths, so grouping
+        # similar-length columns removes intra-wave imbalance even on
+        # uniform shards (each thread still owns its column's write —
+        # bitwise deterministic under any visit order)
+        order = torch.argsort(counts, descending=True,
+                              stable=True).to(torch.int32).contiguous()
         heavy = counts > self.CSC_HEAVY_T
         if not bool(heavy.any()):
-            return None
+            empty = torch.zeros(0, dtype=torch.int32, device=dev)
+            return {
+                "heavy_T": self.CSC_HEAVY_T, "S": self.CSC_TASK_S,
+                "cols": empty,
+                "taskptr": torch.zeros(1, dtype=torch.int32, device=dev),
+                "task_idx": empty, "partial": None, "order": order,
+            }
         cols = torch.nonzero(heavy, as_tuple=False).reshape(-1)
         ntasks = (counts[cols] + self.CSC_TASK_S - 1) // self.CSC_TASK_S
         taskptr = torch.zeros(cols.numel() + 1, dtype=torch.int64,
@@ -183,7 +197,6 @@ class CSRShard:
         torch.cumsum(ntasks, 0, out=taskptr[1:])
         task_idx = torch.repeat_interleave(
             torch.arange(cols.numel(), device=cols.device), ntasks)
-        dev = self.val.device
         return {
             "heavy_T": self.CSC_HEAVY_T,
             "S": self.CSC_TASK_S,
@@ -192,6 +205,7 @@ class CSRShard:
             "task_idx": task_idx.to(torch.int32).contiguous(),
             "partial": torch.empty(int(taskptr[-1]), dtype=torch.float32,
                                    device=dev),
+            "order": order,
         }
 
     def eval(self, w: torch.Tensor, loss_type: int, mask: Optional[torch.Tensor] = None,

@@ -82,9 +82,9 @@ def load() -> ctypes.CDLL:
     lib.agd_csr_margins_multi.argtypes = [P, P, P, P, LL, I, I, P, P]
     lib.agd_csc_grad_multi.restype = I
     lib.agd_csc_grad_multi.argtypes = [P, P, P, P, LL, I, P,
-                                       I, P, P, P, LL, LL, I, P, P]
+                                       I, P, P, P, LL, LL, I, P, P, P]
     lib.agd_csc_grad_skew.restype = I
-    lib.agd_csc_grad_skew.argtypes = [P, P, P, P, LL, I, P, P, P, LL, LL, I, P, P, P]
+    lib.agd_csc_grad_skew.argtypes = [P, P, P, P, LL, I, P, P, P, LL, LL, I, P, P, P, P]
     lib.agd_gram_mult_affine.restype = I
     lib.agd_gram_mult_affine.argtypes = [P, P, D, D, P, P, I, LL, P, P, P, P]
     lib.agd_gram_state_update.restype = I
@@ -320,7 +320,8 @@ def _csc_grad_skew(csc, csc_heavy, mult: torch.Tensor, d: int) -> torch.Tensor:
         int(csc_heavy["heavy_T"]), _ptr(csc_heavy["cols"]),
         _ptr(csc_heavy["taskptr"]), _ptr(csc_heavy["task_idx"]),
         csc_heavy["cols"].numel(), csc_heavy["task_idx"].numel(),
-        int(csc_heavy["S"]), _ptr(csc_heavy["partial"]), _ptr(grad),
+        int(csc_heavy["S"]), _ptr(csc_heavy["partial"]),
+        _ptr(csc_heavy.get("order")), _ptr(grad),
         _stream(cval))
     _check(rc)
     return grad
@@ -516,20 +517,20 @@ def csc_grad_multi(colptr, row, cval, M: torch.Tensor, d: int,
     grad = torch.empty(d * kc, dtype=torch.float32, device=cval.device)
     if csc_heavy is not None:
         n_tasks = csc_heavy["task_idx"].numel()
-        partial = torch.empty(n_tasks * kc, dtype=torch.float32,
-                              device=cval.device)
+        partial = (torch.empty(n_tasks * kc, dtype=torch.float32,
+                               device=cval.device) if n_tasks > 0 else None)
         rc = lib.agd_csc_grad_multi(
             _ptr(colptr.contiguous()), _ptr(row.contiguous()),
             _ptr(cval.contiguous()), _ptr(M.contiguous()), d, kc, _ptr(grad),
             int(csc_heavy["heavy_T"]), _ptr(csc_heavy["cols"]),
             _ptr(csc_heavy["taskptr"]), _ptr(csc_heavy["task_idx"]),
             csc_heavy["cols"].numel(), n_tasks, int(csc_heavy["S"]),
-            _ptr(partial), _stream(cval))
+            _ptr(partial), _ptr(csc_heavy.get("order")), _stream(cval))
     else:
         rc = lib.agd_csc_grad_multi(
             _ptr(colptr.contiguous()), _ptr(row.contiguous()),
             _ptr(cval.contiguous()), _ptr(M.contiguous()), d, kc, _ptr(grad),
-            0, None, None, None, 0, 0, 0, None, _stream(cval))
+            0, None, None, None, 0, 0, 0, None, None, _stream(cval))
     _check(rc)
     return grad
 

@@ -539,10 +539,11 @@ def test_csc_grad_skew_matches_oracle(monkeypatch):
     monkeypatch.setattr(CSRShard, "CSC_HEAVY_T", 2048)
     monkeypatch.setattr(CSRShard, "CSC_TASK_S", 2048)
     shard2 = CSRShard(rowptr, col, val, labels, d)
-    assert shard2.csc_heavy is not None  # hottest col >> 2048 here too
+    assert shard2.csc_heavy["task_idx"].numel() > 0  # hottest col >> 2048
     monkeypatch.setattr(CSRShard, "CSC_HEAVY_T", 10**9)
     shard3 = CSRShard(rowptr, col, val, labels, d)
-    assert shard3.csc_heavy is None
+    # no heavy columns: light-only path (with the sorted visit order)
+    assert shard3.csc_heavy["task_idx"].numel() == 0
     g3, l3 = shard3.eval(w, 0)
     torch.testing.assert_close(gh, g3, rtol=1e-4, atol=1e-4)
 
