@@ -174,13 +174,12 @@ class CSRShard:
         colptr = self.csc[0].to(torch.int64)
         counts = torch.diff(colptr)
         dev = self.val.device
-        # length-sorted visit order for the light kernel: a wave executes
-        # the MAX of its 64 threads' column lengths, so grouping
-        # similar-length columns removes intra-wave imbalance even on
-        # uniform shards (each thread still owns its column's write —
-        # bitwise deterministic under any visit order)
-        order = torch.argsort(counts, descending=True,
-                              stable=True).to(torch.int32).contiguous()
+        # NOTE a length-sorted visit order for the light kernel (to remove
+        # intra-wave imbalance) was measured and REJECTED: adjacent threads
+        # then read SCATTERED csc windows, losing the contiguous-window
+        # coalescing that dominates (uniform 2.26 -> 3.99 ms/step;
+        # profiles/r02_csr_skew_ab.txt). order stays None (identity).
+        order = None
         heavy = counts > self.CSC_HEAVY_T
         if not bool(heavy.any()):
             empty = torch.zeros(0, dtype=torch.int32, device=dev)
