@@ -167,8 +167,8 @@ class CSRShard:
     #: kernel's wave executes the MAX of its 64 threads' column lengths, so
     #: a lower threshold moves imbalance into the wave-parallel heavy path;
     #: env overrides allow threshold A/Bs (profiles/r02_csr_skew_ab.txt).
-    CSC_HEAVY_T = int(__import__("os").environ.get("SPARKAGD_CSC_HEAVY_T", "512"))
-    CSC_TASK_S = int(__import__("os").environ.get("SPARKAGD_CSC_TASK_S", "512"))
+    CSC_HEAVY_T = int(__import__("os").environ.get("SPARKAGD_CSC_HEAVY_T", "16"))
+    CSC_TASK_S = int(__import__("os").environ.get("SPARKAGD_CSC_TASK_S", "128"))
 
     def _build_csc_heavy(self):
         colptr = self.csc[0].to(torch.int64)
