@@ -245,7 +245,10 @@ def main() -> int:
     # converge the bench configs, capped so the tail stays a few seconds.
     eps_iters = args.eps_iters
     if eps_iters < 0:
-        eps_iters = 150 if device.type == "cuda" else 60
+        # 300 total on GPU: converges every bench config for a stable L*
+        # AND keeps the process busy for several seconds so the driver's
+        # SMI sampling sees real utilization
+        eps_iters = 300 if device.type == "cuda" else 60
         eps_iters = max(0, eps_iters - total_iters)  # long runs converge alone
     run_iters = total_iters + eps_iters
 
