@@ -155,12 +155,18 @@ class CSRShard:
         rows = torch.repeat_interleave(
             torch.arange(n, device=self.col.device, dtype=torch.int32), counts
         )
-        order = torch.argsort(self.col.to(torch.int64), stable=True)
+        col64 = self.col.to(torch.int64)
+        order = torch.argsort(col64, stable=True)
+        sorted_col = col64[order]
         csc_row = rows[order].contiguous()
         csc_val = self.val[order].contiguous()
-        colcounts = torch.bincount(self.col.to(torch.int64), minlength=self._d)
-        colptr = torch.zeros(self._d + 1, dtype=torch.int64, device=self.col.device)
-        torch.cumsum(colcounts, dim=0, out=colptr[1:])
+        # colptr straight from the sorted columns (bincount over 64M nnz
+        # measured 90 ms on GPU; searchsorted on the already-sorted array
+        # is a fraction of that)
+        colptr = torch.searchsorted(
+            sorted_col,
+            torch.arange(self._d + 1, device=sorted_col.device,
+                         dtype=torch.int64))
         return colptr.to(torch.int32).contiguous(), csc_row, csc_val
 
     #: heavy-column split parameters (see csc_heavy above). The light
