@@ -112,3 +112,34 @@ def test_streamed_gpu_matches_hbm():
     wr, h = run(streamed, LogisticGradient(), SimpleUpdater(), 1e-10, 5, 0.0,
                 w0, 1.0, math.inf, 0.5, 0.9, True)
     assert h[-1] < h[0] and all(math.isfinite(x) for x in h)
+
+
+@pytest.mark.gpu
+def test_streamed_back_to_back_passes_no_corruption():
+    """Regression for the advisor-flagged double-buffer race: two
+    back-to-back feature passes with NO intervening host sync (margins(x)
+    then margins(z), exactly the optimizer's refresh pattern) must not let
+    the second pass's H2D copies overwrite chunks the first pass's tail
+    kernels still read. Repeat several times — the race was timing
+    dependent."""
+    import torch as _t
+
+    from sparkagd_amd.data import DenseShard
+    from sparkagd_amd.streaming import HostStreamedDenseShard
+
+    g = _t.Generator().manual_seed(41)
+    n, d = 16384, 2048
+    feats = _t.randn((n, d), generator=g)
+    labels = (_t.rand(n, generator=g) < 0.5).float()
+    dev_shard = DenseShard(feats.to("cuda:0"), labels.to("cuda:0"))
+    st = HostStreamedDenseShard(feats, labels, device="cuda:0",
+                                chunk_rows=1024)
+    x = _t.randn(d, generator=g).to("cuda:0")
+    z = _t.randn(d, generator=g).to("cuda:0")
+    ref_x = dev_shard.margins(x)
+    ref_z = dev_shard.margins(z)
+    for _ in range(6):
+        mx = st.margins(x)   # no torch.cuda.synchronize between these
+        mz = st.margins(z)
+        _t.testing.assert_close(mx, ref_x, rtol=1e-4, atol=1e-4)
+        _t.testing.assert_close(mz, ref_z, rtol=1e-4, atol=1e-4)
