@@ -319,10 +319,9 @@ def run_gram(
     # updates in ONE pass — ~12 small launches collapse to ~5 around the
     # dominant K·m stream.
     use_fused = (
-        ncols == 1
-        and dev.type == "cuda"
+        dev.type == "cuda"
         and acc == torch.float32
-        and getattr(gradient, "LOSS_TYPE", -1) >= 0
+        and (getattr(gradient, "LOSS_TYPE", -1) >= 0 or ncols > 1)
         and updater.PROX_KIND in (ops.PROX_SIMPLE, ops.PROX_SQUARED_L2)
         and ops._use_hip(data.features)
     )
@@ -340,16 +339,23 @@ def run_gram(
         nonlocal T
         if T >= max_basis:
             raise RuntimeError("gram basis overflow — raise max_basis")
-        m_y = torch.empty(flat_n, dtype=acc, device=dev)
-        lc_y = torch.empty(2, dtype=torch.float64, device=dev)
-        _hl.gram_mult_affine(xm_o, zm_o, 1.0 - th, th, labels_f32,
-                             gradient.LOSS_TYPE, sw, m_y, lc_y)
-        comm.allreduce_(lc_y)
-        # the fused path bypasses gradient.multiplier_loss, so notify any
-        # counting wrapper (bench.py) that a full loss evaluation happened
-        count_eval = getattr(gradient, "count_eval", None)
-        if count_eval is not None:
-            count_eval()
+        count_eval = None
+        if ncols == 1:
+            m_y = torch.empty(flat_n, dtype=acc, device=dev)
+            lc_y = torch.empty(2, dtype=torch.float64, device=dev)
+            _hl.gram_mult_affine(xm_o, zm_o, 1.0 - th, th, labels_f32,
+                                 gradient.LOSS_TYPE, sw, m_y, lc_y)
+            comm.allreduce_(lc_y)
+            # the fused path bypasses gradient.multiplier_loss, so notify a
+            # counting wrapper (bench.py) that a loss evaluation happened
+            count_eval = getattr(gradient, "count_eval", None)
+            if count_eval is not None:
+                count_eval()
+        else:
+            # multiclass: the softmax multiplier kernel needs materialized
+            # margins; only the state-update fusion applies
+            ym = ops.axpby(1.0 - th, xm_o, th, zm_o)
+            lc_y, m_y = eval_loss_async(ym)
         m_global = op.all_gather_m(m_y)
         gm_raw = op.matvec(m_global)  # unscaled K·m; state kernel applies 1/c
         t = T + 1
@@ -365,15 +371,20 @@ def run_gram(
         T += 1
         cz_t = prox_coeff(cz_o, t, step)
         cx_t = (1.0 - th) * cx_o + th * cz_t
+        lc_x = None
         if backtrack_simple and beta < 1.0:
-            # loss-only x-eval through the same kernel (a=1, b=0); m_y's
-            # buffer is free again (already persisted into Mstore/md)
-            _hl.gram_mult_affine(xm_new, xm_new, 1.0, 0.0, labels_f32,
-                                 gradient.LOSS_TYPE, sw, m_y, lc_x_buf)
-            comm.allreduce_(lc_x_buf)
-            if count_eval is not None:
-                count_eval()
-        return t, step, cz_t, cx_t, zm_new, xm_new, lc_y, row
+            if ncols == 1:
+                # loss-only x-eval through the same kernel (a=1, b=0); m_y's
+                # buffer is free again (already persisted into Mstore/md)
+                _hl.gram_mult_affine(xm_new, xm_new, 1.0, 0.0, labels_f32,
+                                     gradient.LOSS_TYPE, sw, m_y, lc_x_buf)
+                comm.allreduce_(lc_x_buf)
+                if count_eval is not None:
+                    count_eval()
+                lc_x = lc_x_buf
+            else:
+                lc_x, _ = eval_loss_async(xm_new)
+        return t, step, cz_t, cx_t, zm_new, xm_new, lc_y, row, lc_x
 
     def prox_coeff(cz_old: np.ndarray, t_idx: int, step: float) -> np.ndarray:
         out = cz_old.copy()
@@ -407,10 +418,8 @@ def run_gram(
             cy = (1.0 - theta) * cx_old + theta * cz_old
             lc_x = None
             if use_fused:
-                (t_y, step, cz, cx, zm, xm, lc_y, row_y) = fused_y_trial(
+                (t_y, step, cz, cx, zm, xm, lc_y, row_y, lc_x) = fused_y_trial(
                     theta, L, xm_old, zm_old, cz_old, cx_old)
-                if backtrack_simple and beta < 1.0:
-                    lc_x = lc_x_buf
             else:
                 ym = ops.axpby(1.0 - theta, xm_old, theta, zm_old)
                 # Enqueue the whole trial's device work before the first host
