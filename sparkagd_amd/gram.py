@@ -356,8 +356,8 @@ def run_gram(
             # margins; only the state-update fusion applies
             ym = ops.axpby(1.0 - th, xm_o, th, zm_o)
             lc_y, m_y = eval_loss_async(ym)
-        m_global = op.all_gather_m(m_y)
-        gm_raw = op.matvec(m_global)  # unscaled K·m; state kernel applies 1/c
+        m_global = op.all_gather_m(m_y, ncols)
+        gm_raw = op.matvec(m_global, ncols)  # unscaled K·m; state kernel scales by 1/c
         t = T + 1
         step = 1.0 / (th * L_now)
         pz = (1.0 - step * reg_param
