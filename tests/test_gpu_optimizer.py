@@ -308,3 +308,24 @@ def test_gram_multiclass_matches_direct_gpu():
     num = float(torch.norm(w_g - w_d))
     den = float(torch.norm(w_d)) + 1e-30
     assert num / den < 2e-2, (num, den)
+
+
+def test_gram_alternate_backtracking_fused_gpu():
+    """Alternate backtracking (backtrack_tol=inf) through the FUSED gram
+    trial on GPU: fused y-registration + generic x-registration must keep
+    the coefficient-space bookkeeping consistent (T/XB/Mstore/G)."""
+    shard, _ = generate_dense_problem(n=4096, d=32768, seed=83, device=DEV,
+                                      dtype=torch.bfloat16)
+    w0 = torch.zeros(32768, device=DEV, dtype=torch.float32)
+    args = (shard, LogisticGradient(), SquaredL2Updater(), 1e-12, 8, 0.05, w0,
+            1.0, math.inf, 0.5, 0.9, True)
+    w_d, h_d = run(*args, loss_history_mode="backtrack",
+                   backtrack_tol=math.inf)
+    w_g, h_g = run(*args, solver="gram", loss_history_mode="backtrack",
+                   backtrack_tol=math.inf)
+    assert len(h_d) == len(h_g)
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 5e-3 * max(1.0, abs(b)), (a, b)
+    num = float(torch.norm(w_g - w_d))
+    den = float(torch.norm(w_d)) + 1e-30
+    assert num / den < 2e-2, (num, den)

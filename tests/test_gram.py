@@ -126,3 +126,21 @@ def test_gram_checkpoint_resume_explicit_error():
         run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 3, 0.1, w0,
             1.0, math.inf, 0.5, 0.9, True, solver="gram",
             resume_from="/tmp/does_not_exist.safetensors")
+
+
+def test_gram_alternate_backtracking_matches_direct():
+    """backtrack_tol=inf forces the ALTERNATE backtracking test from trial 2
+    on — in the fused GPU path this mixes the fused y-registration with the
+    generic x-basis registration, so the bookkeeping (T, XB, Mstore, G)
+    must stay consistent across both."""
+    data = generate_logistic_data(2.0, -1.5, 3000, seed=77)
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+    args = (data, LogisticGradient(), SquaredL2Updater(), 1e-12, 8, 0.1, w0,
+            1.0, math.inf, 0.5, 0.9, True)
+    w_d, h_d = run(*args, backtrack_tol=math.inf)
+    w_g, h_g = run(*args, solver="gram", backtrack_tol=math.inf)
+    assert len(h_d) == len(h_g)
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 1e-7 * max(1.0, abs(b)), (a, b)
+    torch.testing.assert_close(torch.as_tensor(w_g), torch.as_tensor(w_d),
+                               rtol=1e-6, atol=1e-9)
