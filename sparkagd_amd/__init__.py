@@ -46,7 +46,7 @@ from .parallel.comm import Communicator
 from . import evaluation
 from .streaming import HostStreamedDenseShard
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 __all__ = [
     "AGDConfig",
