@@ -142,6 +142,9 @@ def main() -> int:
     p.add_argument("--csr-cluster", action="store_true",
                    help="column-frequency clustering preprocessing "
                         "(reindex_columns) before training")
+    p.add_argument("--mixed", action="store_true",
+                   help="heterogeneous shard: half the rows dense, half "
+                        "CSR-sparse, over one feature space (MixedShard)")
     p.add_argument("--classes", type=int, default=0,
                    help=">0: multinomial softmax regression with K classes")
     p.add_argument("--eps", type=float, default=1e-3,
@@ -173,6 +176,8 @@ def main() -> int:
     loss_type, grad_cls = LOSSES[args.loss]
 
     t_gen0 = time.perf_counter()
+    if args.mixed and (args.classes > 0 or args.csr):
+        raise SystemExit("--mixed is a binary dense+CSR configuration")
     if args.classes > 0 and args.csr:
         from sparkagd_amd.data import generate_multiclass_csr_problem
 
@@ -197,6 +202,20 @@ def main() -> int:
             from sparkagd_amd.data import reindex_columns
 
             shard, _perm = reindex_columns(shard)
+    elif args.mixed:
+        from sparkagd_amd.data import MixedShard
+
+        half = max(args.rows // 2, 1)
+        dense_part, _w_true = generate_dense_problem(
+            half, args.d, seed=1234 + rank * 7, loss_type=loss_type,
+            device=device, dtype=dtype, label_noise=args.label_noise,
+        )
+        csr_part, _ = generate_csr_problem(
+            args.rows - half, args.d, args.nnz_per_row,
+            seed=4321 + rank * 7, loss_type=loss_type, device=device,
+            col_dist=args.csr_dist, zipf_a=args.csr_zipf_a,
+        )
+        shard = MixedShard([dense_part, csr_part])
     else:
         shard, _w_true = generate_dense_problem(
             args.rows, args.d, seed=1234 + rank * 7, loss_type=loss_type,
@@ -207,7 +226,7 @@ def main() -> int:
                          "dense n×n and the CSR configs have n ≥ 1e6 — "
                          "see sparkagd_amd/gram.py)")
     if args.streamed:
-        if args.classes > 0 or args.csr:
+        if args.classes > 0 or args.csr or args.mixed:
             raise SystemExit("--streamed supports the dense binary configs")
         from sparkagd_amd import HostStreamedDenseShard
 
@@ -350,7 +369,7 @@ def main() -> int:
             "config": {
                 "model": (f"{'csr_' if args.csr else ''}multinomial{args.classes}_regression"
                           if args.classes > 0
-                          else f"{'csr' if args.csr else ('streamed' if args.streamed else 'dense')}_{args.loss}_regression"),
+                          else f"{'csr' if args.csr else ('streamed' if args.streamed else ('mixed' if args.mixed else 'dense'))}_{args.loss}_regression"),
                 "d": args.d,
                 "rows_per_gpu": args.rows,
                 "global_rows": global_rows,
