@@ -755,37 +755,57 @@ __global__ __launch_bounds__(BLOCK) void k_multiplier_multi_anyk(
     const unsigned char* __restrict__ mask,
     const float* __restrict__ sample_weight, ll n, int K, int KC,
     float* __restrict__ M, double* __restrict__ red_part) {
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
   const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
   double lsum = 0.0, cnt = 0.0;
+  // KC is 4-aligned (padded_k), so 16-B lane loads cover the row exactly;
+  // per-element k < K predication handles the logical/padded boundary.
   for (ll i = wave_gid; i < n; i += n_waves) {
     const float* __restrict__ zrow = Z + i * KC;
     float* __restrict__ mrow = M + i * KC;
     if (mask && !mask[i]) {
-      for (int k = lane; k < KC; k += WAVE) mrow[k] = 0.f;
+      for (int k4 = lane * 4; k4 < KC; k4 += WAVE * 4)
+        *(f32x4*)&mrow[k4] = f32x4{0.f, 0.f, 0.f, 0.f};
       continue;
     }
     const int y = (int)labels[i];
     float zmax = -3.0e38f;
-    for (int k = lane; k < K; k += WAVE) zmax = fmaxf(zmax, zrow[k]);
+    for (int k4 = lane * 4; k4 < KC; k4 += WAVE * 4) {
+      const f32x4 zv = *(const f32x4*)&zrow[k4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (k4 + j < K) zmax = fmaxf(zmax, zv[j]);
+    }
     zmax = wave_reduce_max(zmax);
     float esum = 0.f;
-    for (int k = lane; k < K; k += WAVE) esum += __expf(zrow[k] - zmax);
+    for (int k4 = lane * 4; k4 < KC; k4 += WAVE * 4) {
+      const f32x4 zv = *(const f32x4*)&zrow[k4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (k4 + j < K) esum += __expf(zv[j] - zmax);
+    }
     esum = wave_reduce_sum(esum);
     const float inv = 1.0f / esum;
     const float scale = sample_weight ? sample_weight[i] : 1.0f;
     float zy = 0.f;  // only the lane iteration with k == y contributes
-    for (int k = lane; k < KC; k += WAVE) {
-      if (k < K) {
-        const float zk = zrow[k];
-        const float m = __expf(zk - zmax) * inv - ((k == y) ? 1.0f : 0.0f);
-        mrow[k] = m * scale;
-        if (k == y) zy = zk;
-      } else {
-        mrow[k] = 0.f;
+    for (int k4 = lane * 4; k4 < KC; k4 += WAVE * 4) {
+      const f32x4 zv = *(const f32x4*)&zrow[k4];
+      f32x4 mv;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int k = k4 + j;
+        if (k < K) {
+          mv[j] = (__expf(zv[j] - zmax) * inv - ((k == y) ? 1.0f : 0.0f))
+                  * scale;
+          if (k == y) zy = zv[j];
+        } else {
+          mv[j] = 0.f;
+        }
       }
+      *(f32x4*)&mrow[k4] = mv;
     }
     zy = wave_reduce_sum(zy);
     if (lane == 0) {
@@ -1031,6 +1051,42 @@ __global__ __launch_bounds__(BLOCK) void k_axpby(double a, const T* __restrict__
   }
   const ll t = nv * VE + gid;  // tail (< VE elements)
   if (t < n) out[t] = ta * x[t] + tb * y[t];
+}
+
+// Fused AT margin update (direct tracked path): zm' = pz*zm + pg*gm and
+// xm' = (1-theta)*xm + theta*zm' in ONE pass — replaces the prox-margins +
+// interpolation axpby pair (3 reads + 2 writes instead of 4 + 2, one
+// launch instead of two). Matters when the margin state is large
+// (multiclass n*KC flats); algebra identical to the axpby composition.
+template <typename T>
+__global__ __launch_bounds__(BLOCK) void k_at_margin_update(
+    const T* __restrict__ zm_old, const T* __restrict__ xm_old,
+    const T* __restrict__ gm, double pz, double pg, double theta, ll n,
+    T* __restrict__ zm_new, T* __restrict__ xm_new) {
+  constexpr int VE = 16 / sizeof(T);
+  const T tpz = (T)pz, tpg = (T)pg, tth = (T)theta, tom = (T)(1.0 - theta);
+  const ll stride = (ll)gridDim.x * BLOCK;
+  const ll gid = (ll)blockIdx.x * BLOCK + threadIdx.x;
+  const ll nv = n / VE;
+  for (ll i = gid; i < nv; i += stride) {
+    T zo[VE], xo[VE], gv[VE], zn[VE], xn[VE];
+    loadAcc<T, VE>(zm_old + i * VE, zo);
+    loadAcc<T, VE>(xm_old + i * VE, xo);
+    loadAcc<T, VE>(gm + i * VE, gv);
+#pragma unroll
+    for (int k = 0; k < VE; ++k) {
+      zn[k] = tpz * zo[k] + tpg * gv[k];
+      xn[k] = tom * xo[k] + tth * zn[k];
+    }
+    storeAcc<T, VE>(zm_new + i * VE, zn);
+    storeAcc<T, VE>(xm_new + i * VE, xn);
+  }
+  const ll t = nv * VE + gid;  // tail (< VE elements)
+  if (t < n) {
+    const T z = tpz * zm_old[t] + tpg * gm[t];
+    zm_new[t] = z;
+    xm_new[t] = tom * xm_old[t] + tth * z;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1995,6 +2051,30 @@ extern "C" int agd_csc_grad_multi(const void* colptr, const void* row,
       return 2;
   }
 #undef LAUNCH_CG
+  HIP_CHECK(hipGetLastError());
+  return 0;
+}
+
+extern "C" int agd_at_margin_update(const void* zm_old, const void* xm_old,
+                                    const void* gm, double pz, double pg,
+                                    double theta, long long n, int dtype,
+                                    void* zm_new, void* xm_new, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = grid_for(n / (dtype == 2 ? 2 : 4) + 1, BLOCK);
+  if (dtype == 1)
+    hipLaunchKernelGGL((k_at_margin_update<float>), dim3(grid), dim3(BLOCK),
+                       0, s, (const float*)zm_old, (const float*)xm_old,
+                       (const float*)gm, pz, pg, theta, n, (float*)zm_new,
+                       (float*)xm_new);
+  else if (dtype == 2)
+    hipLaunchKernelGGL((k_at_margin_update<double>), dim3(grid), dim3(BLOCK),
+                       0, s, (const double*)zm_old, (const double*)xm_old,
+                       (const double*)gm, pz, pg, theta, n, (double*)zm_new,
+                       (double*)xm_new);
+  else {
+    snprintf(g_err, sizeof(g_err), "at_margin_update: bad dtype %d", dtype);
+    return 2;
+  }
   HIP_CHECK(hipGetLastError());
   return 0;
 }

@@ -39,6 +39,11 @@ class Updater:
         AFFINE_PROX."""
         raise NotImplementedError(f"{type(self).__name__} has no affine prox")
 
+    def prox_margin_coeffs(self, step: float, reg_param: float):
+        """(pz, pg) with prox_margins(wm, gm) = pz*wm + pg*gm; only valid
+        when AFFINE_PROX (used by the fused AT margin-update kernels)."""
+        raise NotImplementedError(f"{type(self).__name__} has no affine prox")
+
     def compute(
         self,
         weights_old: torch.Tensor,
@@ -70,6 +75,9 @@ class SimpleUpdater(Updater):
 
     def prox_margins(self, wm, gm, step, reg_param):
         return ops.axpby(1.0, wm, -step, gm)
+
+    def prox_margin_coeffs(self, step, reg_param):
+        return 1.0, -step
 
 
 class L1Updater(Updater):
@@ -115,3 +123,6 @@ class SquaredL2Updater(Updater):
 
     def prox_margins(self, wm, gm, step, reg_param):
         return ops.axpby(1.0 - step * reg_param, wm, -step, gm)
+
+    def prox_margin_coeffs(self, step, reg_param):
+        return 1.0 - step * reg_param, -step

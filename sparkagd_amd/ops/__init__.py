@@ -162,6 +162,16 @@ def csr_eval_from_margins(rowptr, col, val, margins, labels, loss_type,
                                            sample_weight)
 
 
+def at_margin_update(zm_old, xm_old, gm, pz: float, pg: float, theta: float):
+    """Fused AT margin update: zm' = pz*zm + pg*gm; xm' = (1-th)*xm + th*zm'
+    — one pass on GPU, the axpby composition on CPU (identical algebra)."""
+    if _use_hip(zm_old):
+        return _get_hip().at_margin_update(zm_old, xm_old, gm, pz, pg, theta)
+    zm_new = pz * zm_old + pg * gm
+    xm_new = (1.0 - theta) * xm_old + theta * zm_new
+    return zm_new, xm_new
+
+
 def prox(
     kind: int, w: torch.Tensor, g: torch.Tensor, step: float, lam: float,
     lam2: float = 0.0,

@@ -89,6 +89,8 @@ def load() -> ctypes.CDLL:
     lib.agd_gram_mult_affine.argtypes = [P, P, D, D, P, P, I, LL, P, P, P, P]
     lib.agd_gram_state_update.restype = I
     lib.agd_gram_state_update.argtypes = [P, P, P, P, D, D, D, D, LL, P, P, P, P, P, P]
+    lib.agd_at_margin_update.restype = I
+    lib.agd_at_margin_update.argtypes = [P, P, P, D, D, D, LL, I, P, P, P]
 
     _lib = lib
     return lib
@@ -726,3 +728,19 @@ def gram_state_update(gm_raw: torch.Tensor, m_y: torch.Tensor,
         _ptr(xb_t), _ptr(md), _ptr(mstore_t), _ptr(zm_new), _ptr(xm_new),
         _stream(gm_raw))
     _check(rc)
+
+
+def at_margin_update(zm_old: torch.Tensor, xm_old: torch.Tensor,
+                     gm: torch.Tensor, pz: float, pg: float, theta: float):
+    """(zm_new, xm_new) = (pz*zm + pg*gm, (1-theta)*xm + theta*zm_new) in one
+    fused pass (direct tracked path)."""
+    lib = load()
+    zm_new = torch.empty_like(zm_old)
+    xm_new = torch.empty_like(xm_old)
+    rc = lib.agd_at_margin_update(
+        _ptr(zm_old.contiguous()), _ptr(xm_old.contiguous()),
+        _ptr(gm.contiguous()), float(pz), float(pg), float(theta),
+        zm_old.numel(), _VEC_DTYPE[zm_old.dtype], _ptr(zm_new), _ptr(xm_new),
+        _stream(zm_old))
+    _check(rc)
+    return zm_new, xm_new

@@ -259,8 +259,19 @@ def run(
             x = ops.axpby(1.0 - theta, x_old, theta, z)
             if tracking:
                 gm = gradient.margins(data, g_y)  # the only other data pass
-                zm = updater.prox_margins(zm_old, gm, step, reg_param)
-                xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
+                try:
+                    pz, pg = updater.prox_margin_coeffs(step, reg_param)
+                except NotImplementedError:
+                    pz = None
+                if pz is not None:
+                    # fused: zm' and xm' in one pass (k_at_margin_update) —
+                    # one launch + one fewer read stream than the
+                    # prox-margins + axpby pair; identical algebra
+                    zm, xm = ops.at_margin_update(zm_old, xm_old, gm,
+                                                  pz, pg, theta)
+                else:
+                    zm = updater.prox_margins(zm_old, gm, step, reg_param)
+                    xm = ops.axpby(1.0 - theta, xm_old, theta, zm)
 
             if beta >= 1.0:
                 scal = None  # computed after the loop for the convergence test
