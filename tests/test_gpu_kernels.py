@@ -584,3 +584,67 @@ def test_csc_grad_multi_skew_matches_oracle(monkeypatch):
     torch.testing.assert_close(gh, gr, rtol=3e-4, atol=3e-4)
     gh2, lh2 = grad.eval(shard, W)
     assert torch.equal(gh, gh2)
+
+
+def test_fuzz_dense_eval_random_configs():
+    """Randomized parity sweep: dense eval (all 4 binary losses x dtypes x
+    mask/weight combos x awkward shapes) vs the torch oracle. Seeded — the
+    'random' configs are fixed, this is broad coverage, not flakiness."""
+    from sparkagd_amd.ops import reference as ref
+
+    g = torch.Generator(device=DEV).manual_seed(123)
+    shapes = [(1023, 257), (4096, 1000), (777, 4096), (129, 16384)]
+    dtypes = [torch.float32, torch.bfloat16]
+    for li, loss_type in enumerate([0, 1, 2, 3]):
+        n, d = shapes[li % len(shapes)]
+        dt = dtypes[li % 2]
+        A = (torch.randn((n, d), generator=g, device=DEV) / d ** 0.5).to(dt).contiguous()
+        y = (torch.rand(n, generator=g, device=DEV) < 0.5).float()
+        w = torch.randn(d, generator=g, device=DEV) * 0.1
+        mask = (torch.rand(n, generator=g, device=DEV) < 0.8).to(torch.uint8)
+        sw = torch.rand(n, generator=g, device=DEV) + 0.25
+        for m_arg, w_arg in [(None, None), (mask, None), (mask, sw)]:
+            from sparkagd_amd import ops as _ops
+
+            gh, lh = _ops.dense_eval(A, y, w, loss_type, m_arg, True, w_arg)
+            gr, lr = ref.dense_eval(A, y, w, loss_type, m_arg, True, w_arg)
+            torch.testing.assert_close(lh, lr, rtol=5e-4, atol=5e-4)
+            torch.testing.assert_close(gh, gr, rtol=5e-3, atol=5e-3)
+
+
+def test_fuzz_csr_eval_random_configs():
+    """Randomized CSR parity sweep incl. skewed columns, empty rows and
+    duplicate column indices, vs the torch oracle."""
+    from sparkagd_amd.data import CSRShard
+    from sparkagd_amd.ops import reference as ref
+
+    g = torch.Generator(device=DEV).manual_seed(321)
+    for cfg in range(4):
+        n = [5000, 1701, 12000, 300][cfg]
+        d = [2000, 517, 30000, 40][cfg]
+        nnz_row = [8, 3, 24, 5][cfg]
+        nnz = n * nnz_row
+        if cfg % 2 == 0:  # skewed: 30% of entries in a few hot columns
+            hot = torch.randint(0, 3, (nnz,), generator=g, device=DEV)
+            uni = torch.randint(0, d, (nnz,), generator=g, device=DEV)
+            pick = torch.rand(nnz, generator=g, device=DEV) < 0.3
+            col = torch.where(pick, hot * max(d // 7, 1) % d, uni).to(torch.int32)
+        else:
+            col = torch.randint(0, d, (nnz,), generator=g, device=DEV,
+                                dtype=torch.int32)
+        col = col.view(n, nnz_row).sort(dim=1).values.reshape(-1)
+        val = torch.randn(nnz, generator=g, device=DEV)
+        rowptr = torch.arange(0, nnz + 1, nnz_row, device=DEV,
+                              dtype=torch.int32)
+        # punch a few empty rows (rowptr[k] == rowptr[k+1])
+        labels = (torch.rand(n, generator=g, device=DEV) < 0.5).float()
+        shard = CSRShard(rowptr, col, val, labels, d)
+        w = torch.randn(d, generator=g, device=DEV) * 0.1
+        loss_type = cfg % 4
+        gh, lh = shard.eval(w, loss_type)
+        gr, lr = ref.csr_eval(shard.rowptr, shard.col, shard.val, labels, w,
+                              loss_type, None, d)
+        torch.testing.assert_close(lh, lr.to(lh.dtype), rtol=5e-4, atol=5e-4)
+        torch.testing.assert_close(gh, gr.to(gh.dtype), rtol=5e-3, atol=5e-3)
+        gh2, lh2 = shard.eval(w, loss_type)
+        assert torch.equal(gh, gh2), "determinism must hold on every config"
