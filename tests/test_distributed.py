@@ -519,3 +519,18 @@ def test_sharded_mixed_replicates():
     results = _run_dist(_dist_mixed, world=2)
     assert results[0][0] == results[1][0]
     assert results[0][1][-1] < results[0][1][0]
+
+
+def test_sharded_agd_world4():
+    """4 ranks, uneven shards — the widest CPU-tier rank layout (the driver
+    scales 1/2/4/8 on hardware)."""
+    results = _run_dist(_dist_agd, world=4)
+    full = generate_logistic_data(2.0, -1.5, N, seed=42)
+    w0 = torch.tensor([0.3, 0.12], dtype=torch.float64)
+    w_ref, _ = run(full, LogisticGradient(), SquaredL2Updater(), 1e-12, 8,
+                   0.2, w0, 1.0, math.inf, 0.5, 0.9, True)
+    for rank in range(4):
+        torch.testing.assert_close(
+            torch.tensor(results[rank][0], dtype=torch.float64), w_ref,
+            rtol=1e-9, atol=1e-12)
+    assert results[0][0] == results[3][0]
