@@ -1943,31 +1943,44 @@ __global__ __launch_bounds__(BLOCK) void k_csc_heavy_combine_multi(
   }
 }
 
-// Lane-per-class CSC multiclass gradient (see k_csr_margins_multi): each
-// sub-group of KC lanes owns one column; M-row gathers and grad writes are
-// fully coalesced and no lane owns more than one output element.
+// CSC multiclass gradient, thread-per-column with 16-B M loads/grad
+// stores. A lane-per-class variant (like k_csr_margins_multi) was measured
+// and REJECTED here: the random-gather target M is n*KC ~ 64 MB and
+// LLC-resident, so coalescing its rows buys nothing, while thread-per-
+// column keeps adjacent threads streaming ADJACENT csc val/row windows
+// (1.46 vs 2.10 ms/pass at K=16, d=1e7 — profiles/r02_csr_multi notes).
 template <int KC>
 __global__ __launch_bounds__(BLOCK) void k_csc_grad_multi(
     const int* __restrict__ colptr, const int* __restrict__ row,
     const float* __restrict__ val, const float* __restrict__ M, ll d,
     int heavy_T, const int* __restrict__ order, float* __restrict__ grad) {
-  constexpr int CPW = WAVE / KC;  // columns (sub-groups) per wave
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int sub = lane / KC;
-  const int cls = lane - sub * KC;
-  const int wid = threadIdx.x / WAVE;
-  const ll wave_gid = (ll)blockIdx.x * WAVES_PER_BLOCK + wid;
-  const ll n_waves = (ll)gridDim.x * WAVES_PER_BLOCK;
-  for (ll cbase = wave_gid * CPW; cbase < d; cbase += n_waves * CPW) {
-    const ll i = cbase + sub;
-    if (i >= d) continue;
+  using f32x4 = __attribute__((ext_vector_type(4))) float;
+  const ll stride = (ll)gridDim.x * BLOCK;
+  for (ll i = (ll)blockIdx.x * BLOCK + threadIdx.x; i < d; i += stride) {
     const ll c = order ? (ll)order[i] : i;
     const int k_lo = colptr[c], k_hi = colptr[c + 1];
     if (k_hi - k_lo > heavy_T) continue;  // heavy path owns grad[c]
-    float acc = 0.f;
-    for (int k = k_lo; k < k_hi; ++k)
-      acc += val[k] * M[(ll)row[k] * KC + cls];
-    grad[c * KC + cls] = acc;
+    float acc[KC];
+#pragma unroll
+    for (int j = 0; j < KC; ++j) acc[j] = 0.f;
+    for (int k = k_lo; k < k_hi; ++k) {
+      const float v = val[k];
+      const float* __restrict__ mr = M + (ll)row[k] * KC;
+#pragma unroll
+      for (int ch = 0; ch < KC / 4; ++ch) {
+        const f32x4 mv = *(const f32x4*)(mr + ch * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[ch * 4 + j] += v * mv[j];
+      }
+    }
+    float* __restrict__ gr = grad + c * KC;
+#pragma unroll
+    for (int ch = 0; ch < KC / 4; ++ch) {
+      f32x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) o[j] = acc[ch * 4 + j];
+      *(f32x4*)(gr + ch * 4) = o;
+    }
   }
 }
 
@@ -2020,7 +2033,7 @@ extern "C" int agd_csc_grad_multi(const void* colptr, const void* row,
                                   void* partial, const void* light_order,
                                   void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  const int grid = grid_for((ll)d * kc, BLOCK);  // lane-per-class layout
+  const int grid = grid_for(d, BLOCK);  // thread per column
   const int light_T = (n_tasks > 0) ? heavy_T : 0x7fffffff;
 #define LAUNCH_CG(KCV)                                                         \
   do {                                                                         \
