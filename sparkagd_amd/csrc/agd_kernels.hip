@@ -1869,10 +1869,20 @@ __global__ __launch_bounds__(BLOCK) void k_csr_margins_multi(
     const ll r = rbase + sub;
     if (r >= n) continue;
     const int k_lo = rowptr[r], k_hi = rowptr[r + 1];
-    float acc = 0.f;
-    for (int k = k_lo; k < k_hi; ++k)
-      acc += val[k] * w_elem<TW>(w[(ll)col[k] * KC + cls]);
-    Z[r * KC + cls] = acc;
+    // 4 accumulators: independent W-line gathers in flight per sub-group
+    // (the serial per-row loop is gather-latency-exposed; measured
+    // 6.22 -> 5.72 ms/step at 2 accumulators already)
+    float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+    int k = k_lo;
+    for (; k + 3 < k_hi; k += 4) {
+      a0 += val[k] * w_elem<TW>(w[(ll)col[k] * KC + cls]);
+      a1 += val[k + 1] * w_elem<TW>(w[(ll)col[k + 1] * KC + cls]);
+      a2 += val[k + 2] * w_elem<TW>(w[(ll)col[k + 2] * KC + cls]);
+      a3 += val[k + 3] * w_elem<TW>(w[(ll)col[k + 3] * KC + cls]);
+    }
+    for (; k < k_hi; ++k)
+      a0 += val[k] * w_elem<TW>(w[(ll)col[k] * KC + cls]);
+    Z[r * KC + cls] = (a0 + a1) + (a2 + a3);
   }
 }
 
