@@ -363,7 +363,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": vs_baseline,
-            "dtype": ("f32" if args.csr else args.dtype) if device.type == "cuda" else "f32",
+            "dtype": (("f32" if args.csr else (f"{args.dtype}+f32" if args.mixed else args.dtype)) if device.type == "cuda" else "f32"),
             "data": "synthetic",
             "rows_per_sec": rows_per_sec,
             "config": {
