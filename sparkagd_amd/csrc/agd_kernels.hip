@@ -1570,6 +1570,83 @@ static hipblasLtHandle_t g_lt_handle = nullptr;
 static void* g_lt_ws = nullptr;
 static const size_t LT_WS_BYTES = 64u << 20;
 
+// Measured per-shape algo autotune (round 2): the single heuristic
+// candidate left the K=1000 margins GEMM at ~21% MFMA / ~1 TB/s — neither
+// bound. One-time per (shape, layout-kind): request up to 16 candidates,
+// time each with events on the caller's stream, cache the winner. Runs
+// during warmup (first call per shape); only when beta == 0, so repeated
+// tuning launches cannot corrupt an accumulating C.
+struct LtAlgoEntry {
+  long long m, n, k;
+  int kind;
+  hipblasLtMatmulAlgo_t algo;
+};
+static LtAlgoEntry g_lt_algos[64];
+static int g_lt_algo_count = 0;
+
+static int lt_pick_algo(int kind, hipblasLtMatmulDesc_t op,
+                        hipblasLtMatrixLayout_t la, hipblasLtMatrixLayout_t lb,
+                        hipblasLtMatrixLayout_t lc, const void* Amat,
+                        const void* Bmat, void* Cmat, float beta, long long m,
+                        long long n, long long k, hipStream_t s,
+                        hipblasLtMatmulAlgo_t* out) {
+  for (int i = 0; i < g_lt_algo_count; ++i)
+    if (g_lt_algos[i].kind == kind && g_lt_algos[i].m == m &&
+        g_lt_algos[i].n == n && g_lt_algos[i].k == k) {
+      *out = g_lt_algos[i].algo;
+      return 0;
+    }
+  hipblasLtMatmulPreference_t pref = nullptr;
+  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
+      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &LT_WS_BYTES,
+      sizeof(LT_WS_BYTES)));
+  hipblasLtMatmulHeuristicResult_t heur[16];
+  int found = 0;
+  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(g_lt_handle, op, la, lb, lc, lc,
+                                           pref, 16, heur, &found));
+  hipblasLtMatmulPreferenceDestroy(pref);
+  if (found < 1) {
+    snprintf(g_err, sizeof(g_err), "hipblaslt: no algo for kind %d %lldx%lldx%lld",
+             kind, m, n, k);
+    return 3;
+  }
+  int best = 0;
+  if (found > 1 && beta == 0.0f) {
+    const float alpha = 1.0f;
+    hipEvent_t e0, e1;
+    HIP_CHECK(hipEventCreate(&e0));
+    HIP_CHECK(hipEventCreate(&e1));
+    float best_ms = 1e30f;
+    for (int a = 0; a < found; ++a) {
+      // warm once, then time 2 reps
+      if (hipblasLtMatmul(g_lt_handle, op, &alpha, Bmat, la, Amat, lb, &beta,
+                          Cmat, lc, Cmat, lc, &heur[a].algo, g_lt_ws,
+                          LT_WS_BYTES, s) != HIPBLAS_STATUS_SUCCESS)
+        continue;
+      HIP_CHECK(hipEventRecord(e0, s));
+      for (int r = 0; r < 2; ++r)
+        (void)hipblasLtMatmul(g_lt_handle, op, &alpha, Bmat, la, Amat, lb,
+                              &beta, Cmat, lc, Cmat, lc, &heur[a].algo,
+                              g_lt_ws, LT_WS_BYTES, s);
+      HIP_CHECK(hipEventRecord(e1, s));
+      HIP_CHECK(hipEventSynchronize(e1));
+      float ms = 1e30f;
+      HIP_CHECK(hipEventElapsedTime(&ms, e0, e1));
+      if (ms < best_ms) {
+        best_ms = ms;
+        best = a;
+      }
+    }
+    HIP_CHECK(hipEventDestroy(e0));
+    HIP_CHECK(hipEventDestroy(e1));
+  }
+  *out = heur[best].algo;
+  if (g_lt_algo_count < 64)
+    g_lt_algos[g_lt_algo_count++] = {m, n, k, kind, heur[best].algo};
+  return 0;
+}
+
 extern "C" int agd_gemm_bf16f32_nt(const void* A, const void* B, void* C,
                                    long long m, long long n, long long k,
                                    float beta, void* stream) {
@@ -1592,23 +1669,14 @@ extern "C" int agd_gemm_bf16f32_nt(const void* A, const void* B, void* C,
   LT_CHECK(hipblasLtMatrixLayoutCreate(&lc, HIP_R_32F, n, m, n));
   const float alpha = 1.0f;
 
-  hipblasLtMatmulPreference_t pref = nullptr;
-  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
-  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
-      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &LT_WS_BYTES,
-      sizeof(LT_WS_BYTES)));
-  hipblasLtMatmulHeuristicResult_t heur;
-  int found = 0;
-  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(g_lt_handle, op, la, lb, lc, lc,
-                                           pref, 1, &heur, &found));
-  if (found < 1) {
-    snprintf(g_err, sizeof(g_err), "hipblaslt: no algo for bf16f32 %lldx%lldx%lld",
-             m, n, k);
-    return 3;
+  hipblasLtMatmulAlgo_t algo;
+  {
+    const int rc = lt_pick_algo(0, op, la, lb, lc, A, B, C, beta, m, n, k, s,
+                                &algo);
+    if (rc != 0) return rc;
   }
   LT_CHECK(hipblasLtMatmul(g_lt_handle, op, &alpha, B, la, A, lb, &beta, C, lc,
-                           C, lc, &heur.algo, g_lt_ws, LT_WS_BYTES, s));
-  hipblasLtMatmulPreferenceDestroy(pref);
+                           C, lc, &algo, g_lt_ws, LT_WS_BYTES, s));
   hipblasLtMatrixLayoutDestroy(lc);
   hipblasLtMatrixLayoutDestroy(lb);
   hipblasLtMatrixLayoutDestroy(la);
@@ -1642,23 +1710,14 @@ extern "C" int agd_gemm_bf16f32_tn(const void* A, const void* M, void* C,
   LT_CHECK(hipblasLtMatrixLayoutCreate(&lc, HIP_R_32F, kc, d, kc));
   const float alpha = 1.0f, beta = 0.0f;
 
-  hipblasLtMatmulPreference_t pref = nullptr;
-  LT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
-  LT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
-      pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &LT_WS_BYTES,
-      sizeof(LT_WS_BYTES)));
-  hipblasLtMatmulHeuristicResult_t heur;
-  int found = 0;
-  LT_CHECK(hipblasLtMatmulAlgoGetHeuristic(g_lt_handle, op, la, lb, lc, lc,
-                                           pref, 1, &heur, &found));
-  if (found < 1) {
-    snprintf(g_err, sizeof(g_err), "hipblaslt: no algo for tn %lldx%lldx%lld",
-             d, kc, n);
-    return 3;
+  hipblasLtMatmulAlgo_t algo;
+  {
+    const int rc = lt_pick_algo(1, op, la, lb, lc, A, M, C, beta, d, kc, n, s,
+                                &algo);
+    if (rc != 0) return rc;
   }
   LT_CHECK(hipblasLtMatmul(g_lt_handle, op, &alpha, M, la, A, lb, &beta, C, lc,
-                           C, lc, &heur.algo, g_lt_ws, LT_WS_BYTES, s));
-  hipblasLtMatmulPreferenceDestroy(pref);
+                           C, lc, &algo, g_lt_ws, LT_WS_BYTES, s));
   hipblasLtMatrixLayoutDestroy(lc);
   hipblasLtMatrixLayoutDestroy(lb);
   hipblasLtMatrixLayoutDestroy(la);
