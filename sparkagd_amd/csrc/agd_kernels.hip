@@ -1612,7 +1612,12 @@ static int lt_pick_algo(int kind, hipblasLtMatmulDesc_t op,
     return 3;
   }
   int best = 0;
-  if (found > 1 && beta == 0.0f) {
+  // Tune only shapes whose per-call cost is small: for multi-TFLOP GEMMs
+  // (e.g. the one-time Gram K build at ~1.7e13 FLOPs/chunk) the 3x16
+  // tuning launches cost seconds while the heuristic is already near
+  // roofline — measured 0.46 s -> 2.2 s build regression before this cap.
+  const double tune_flops = 2.0 * (double)m * (double)n * (double)k;
+  if (found > 1 && beta == 0.0f && tune_flops < 1.0e12) {
     const float alpha = 1.0f;
     hipEvent_t e0, e1;
     HIP_CHECK(hipEventCreate(&e0));
