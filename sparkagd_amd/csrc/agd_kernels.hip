@@ -1616,8 +1616,22 @@ static int lt_pick_algo(int kind, hipblasLtMatmulDesc_t op,
   // (e.g. the one-time Gram K build at ~1.7e13 FLOPs/chunk) the 3x16
   // tuning launches cost seconds while the heuristic is already near
   // roofline — measured 0.46 s -> 2.2 s build regression before this cap.
+  // Multi-rank runs (WORLD_SIZE > 1) take the deterministic heuristic[0]
+  // instead: a timing-based pick could differ across ranks, and the
+  // replicated-update design requires bit-identical math on every rank.
+  // SPARKAGD_GEMM_TUNE=0/1 overrides.
+  static int tune_enabled = -1;
+  if (tune_enabled < 0) {
+    const char* e = getenv("SPARKAGD_GEMM_TUNE");
+    if (e != nullptr)
+      tune_enabled = (e[0] != '0');
+    else {
+      const char* ws = getenv("WORLD_SIZE");
+      tune_enabled = !(ws != nullptr && atoll(ws) > 1);
+    }
+  }
   const double tune_flops = 2.0 * (double)m * (double)n * (double)k;
-  if (found > 1 && beta == 0.0f && tune_flops < 1.0e12) {
+  if (tune_enabled && found > 1 && beta == 0.0f && tune_flops < 1.0e12) {
     const float alpha = 1.0f;
     hipEvent_t e0, e1;
     HIP_CHECK(hipEventCreate(&e0));
