@@ -24,6 +24,12 @@ instead of ~66 GB of shard traffic. The trade-offs, stated plainly:
 * per-GPU iteration work grows with world size (K row-block is
   n_local × n_global), so weak scaling of iteration *throughput* is flat —
   absolute time-to-solution still wins whenever n_global ≲ d;
+* the gradient basis grows by one vector per backtracking trial, so the
+  basis stores (Mstore f32 + XB f64: ~12·max_basis·n_local bytes with
+  max_basis = 8·num_iterations+8) bound the horizon — fine for the
+  hundreds-of-iterations solves this regime needs (157 MB at the headline
+  config × 300 iterations); open-ended training belongs to the direct
+  solver;
 * requires a dense shard, an AFFINE prox (Simple/SquaredL2), full-batch
   evaluations (no mini-batch masks), and fp32 (or fp64) accumulation
   identical in class to the direct path. Multiclass (softmax) gradients are
