@@ -118,6 +118,9 @@ class _GLMTrainer:
         initial_weights: Optional[torch.Tensor] = None,
         comm: Optional[Communicator] = None,
         config: Optional[AGDConfig] = None,
+        checkpoint_path: Optional[str] = None,
+        checkpoint_every: int = 0,
+        resume_from: Optional[str] = None,
     ) -> LinearModel:
         cfg = config or AGDConfig()
         cfg.num_iterations = num_iterations
@@ -126,6 +129,9 @@ class _GLMTrainer:
         if updater is None:
             updater = SquaredL2Updater() if reg_param > 0 else SimpleUpdater()
         opt = AcceleratedGradientDescent(cls.GRADIENT_CLS(), updater, cfg, comm)
+        opt.checkpoint_path = checkpoint_path
+        opt.checkpoint_every = checkpoint_every
+        opt.resume_from = resume_from
         if initial_weights is None:
             wdtype = torch.float64 if data.device.type == "cpu" else torch.float32
             initial_weights = torch.zeros(data.d, device=data.device, dtype=wdtype)
