@@ -31,6 +31,8 @@ run_cfg multinomial_k1000        --classes 1000 --dim 2048 --rows 100000 --steps
 run_cfg csr_multinomial_k16      --csr --classes 16 --rows 1000000 --dim 10000000 --steps 10 --warmup 2 $E
 run_cfg mixed_dense_csr          --mixed --rows 32768 --steps 10 --warmup 2 $E
 run_cfg hinge_smooth_l2          --loss hinge --reg 1e-4 --steps 10 --warmup 2 $E
+run_cfg fp8_d1e6                 --dtype f8 --steps 20 --warmup 3 $E
+run_cfg capacity_92gb            --rows 49152 --steps 5 --warmup 1 $E
 run_cfg dense_n10m_d10k          --rows 10000000 --dim 10240 --steps 5 --warmup 1 $E
 run_cfg lsq_186gb                --loss lsq --rows 1000000 --dim 100096 --steps 3 --warmup 1 $E
 echo "wrote $OUT" >&2
