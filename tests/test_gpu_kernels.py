@@ -648,3 +648,26 @@ def test_fuzz_csr_eval_random_configs():
         torch.testing.assert_close(gh, gr.to(gh.dtype), rtol=5e-3, atol=5e-3)
         gh2, lh2 = shard.eval(w, loss_type)
         assert torch.equal(gh, gh2), "determinism must hold on every config"
+
+
+def test_fuzz_multiclass_eval_random_configs():
+    """Randomized multinomial parity sweep: eval_multi vs the torch oracle
+    across K (VALU-templated and generic), dtypes and mask/weight combos."""
+    from sparkagd_amd.ops import multiclass as mc
+
+    g = torch.Generator(device=DEV).manual_seed(777)
+    cases = [(5000, 333, 3, torch.float32), (4096, 512, 16, torch.bfloat16),
+             (2048, 1024, 37, torch.bfloat16), (1000, 257, 64, torch.float32)]
+    for n, d, K, dt in cases:
+        A = (torch.randn((n, d), generator=g, device=DEV) / d ** 0.5).to(dt).contiguous()
+        y = torch.randint(0, K, (n,), generator=g, device=DEV).float()
+        W = (torch.randn(d * K, generator=g, device=DEV) * 0.2).contiguous()
+        mask = (torch.rand(n, generator=g, device=DEV) < 0.75).to(torch.uint8)
+        sw = torch.rand(n, generator=g, device=DEV) + 0.5
+        for m_arg, w_arg in [(None, None), (mask, sw)]:
+            gh, lh = mc.eval_multi(A, y, W, K, m_arg, True, w_arg)
+            gr, lr = mc.ref_eval_multi(A.float(), y, W, K, m_arg, True, w_arg)
+            torch.testing.assert_close(lh, lr, rtol=1e-4, atol=1e-4)
+            torch.testing.assert_close(gh, gr, rtol=5e-3, atol=5e-3)
+            gh2, lh2 = mc.eval_multi(A, y, W, K, m_arg, True, w_arg)
+            assert torch.equal(gh, gh2) and torch.equal(lh, lh2)

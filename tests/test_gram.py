@@ -144,3 +144,33 @@ def test_gram_alternate_backtracking_matches_direct():
         assert abs(a - b) < 1e-7 * max(1.0, abs(b)), (a, b)
     torch.testing.assert_close(torch.as_tensor(w_g), torch.as_tensor(w_d),
                                rtol=1e-6, atol=1e-9)
+
+
+@pytest.mark.parametrize("seed", [101, 202, 303])
+def test_gram_matches_direct_randomized(seed):
+    """Property sweep: random problem shapes/regs/losses — the Gram solver's
+    trajectory equals the direct solver's (f64 CPU, tight tolerance)."""
+    import numpy as _np
+
+    rng = _np.random.RandomState(seed)
+    n = int(rng.randint(300, 1500))
+    d = int(rng.randint(3, 40))
+    reg = float(rng.choice([0.0, 0.01, 0.3]))
+    iters = int(rng.randint(4, 12))
+    torch.manual_seed(seed)
+    feats = torch.randn(n, d, dtype=torch.float64) / math.sqrt(d)
+    labels = (torch.rand(n, dtype=torch.float64) < 0.5).double()
+    from sparkagd_amd.data import DenseShard
+
+    data = DenseShard(feats, labels)
+    w0 = torch.randn(d, dtype=torch.float64) * 0.1
+    upd = SquaredL2Updater() if reg > 0 else SimpleUpdater()
+    w_d, h_d = run(data, LogisticGradient(), upd, 1e-14, iters, reg, w0,
+                   1.0, math.inf, 0.5, 0.9, True)
+    w_g, h_g = run(data, LogisticGradient(), upd, 1e-14, iters, reg, w0,
+                   1.0, math.inf, 0.5, 0.9, True, solver="gram")
+    assert len(h_d) == len(h_g)
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 1e-7 * max(1.0, abs(b)), (seed, a, b)
+    torch.testing.assert_close(torch.as_tensor(w_g), torch.as_tensor(w_d),
+                               rtol=1e-6, atol=1e-9)
