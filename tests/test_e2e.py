@@ -80,3 +80,27 @@ def test_example_scripts_cpu():
     )
     assert out.returncode == 0, out.stderr[-2000:]
     assert "save/load round-trip: ok" in out.stdout
+
+
+def test_baseline_config1_plumbing():
+    """BASELINE.json config #1 at its exact shape: dense logistic n=1k,
+    d=100, CPU — the reference's `local[2]` plumbing tier. AGD reaches GD's
+    loss with far fewer iterations (the suite's iteration-advantage
+    contract, Suite.scala:60-90)."""
+    import math
+
+    import torch
+
+    from sparkagd_amd import (LogisticGradient, SimpleUpdater, run,
+                              run_mini_batch)
+    from sparkagd_amd.data import generate_dense_problem
+
+    shard, _ = generate_dense_problem(1000, 100, seed=12, dtype=torch.float64)
+    w0 = torch.zeros(100, dtype=torch.float64)
+    w_agd, h_agd = run(shard, LogisticGradient(), SimpleUpdater(), 1e-12, 10,
+                       0.0, w0, 1.0, math.inf, 0.5, 0.9, True)
+    w_gd, h_gd = run_mini_batch(shard, LogisticGradient(), SimpleUpdater(),
+                                1.0, 50, 0.0, 1.0, w0)
+    assert len(h_agd) == 10 and len(h_gd) == 50
+    # AGD(10) at or below GD(50) within the suite's 2% relTol
+    assert h_agd[-1] <= h_gd[-1] * 1.02, (h_agd[-1], h_gd[-1])
