@@ -245,3 +245,49 @@ def _cfg_no_tracking():
     # bitwise resume requires the non-tracking path (tracked margins are
     # recomputed on resume — equivalent, not bitwise)
     return AGDConfig(track_margins=False)
+
+
+def test_error_paths_are_loud():
+    """User errors fail loudly with actionable messages (the no-silent-
+    fallback policy): index-range guard, shape checks, config validation,
+    solver/mode typos."""
+    import pytest
+
+    from sparkagd_amd import AGDConfig, LogisticGradient, SimpleUpdater, run
+    from sparkagd_amd.data import CSRShard, DenseShard
+
+    # CSR int32 index-range guard (round 2)
+    rp = torch.tensor([0, 1], dtype=torch.int64)
+    col = torch.tensor([0], dtype=torch.int32)
+    val = torch.ones(1)
+    with pytest.raises(ValueError, match="int32 index range"):
+        CSRShard(rp, col, val, torch.zeros(1), d=2**31)
+
+    # shape checks
+    with pytest.raises(ValueError, match=r"\[n, d\]"):
+        DenseShard(torch.zeros(4), torch.zeros(4))
+    with pytest.raises(ValueError, match=r"labels"):
+        DenseShard(torch.zeros(4, 2), torch.zeros(3))
+
+    # config validation
+    with pytest.raises(ValueError, match="alpha"):
+        AGDConfig(alpha=1.5).validate()
+    with pytest.raises(ValueError, match="loss_history_mode"):
+        AGDConfig(loss_history_mode="bogus").validate()
+    with pytest.raises(ValueError, match="solver"):
+        AGDConfig(solver="bogus").validate()
+
+    # run() solver typo
+    data = generate_logistic_data(2.0, -1.5, 100, seed=1)
+    with pytest.raises(ValueError, match="direct"):
+        run(data, LogisticGradient(), SimpleUpdater(), 0.0, 2, 0.0,
+            torch.zeros(2, dtype=torch.float64), 1.0, float("inf"), 0.5,
+            0.9, True, solver="bogus")
+
+    # gram demands an affine prox
+    from sparkagd_amd.models.updater import L1Updater
+
+    with pytest.raises(ValueError, match="affine prox"):
+        run(data, LogisticGradient(), L1Updater(), 0.0, 2, 0.1,
+            torch.zeros(2, dtype=torch.float64), 1.0, float("inf"), 0.5,
+            0.9, True, solver="gram")
