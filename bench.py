@@ -377,7 +377,7 @@ def main() -> int:
                 "solver": args.solver,
                 "reg_param": args.reg,
                 "label_noise": args.label_noise,
-                "csr_dist": args.csr_dist if args.csr else None,
+                "csr_dist": args.csr_dist if (args.csr or args.mixed) else None,
                 "csr_cluster": bool(args.csr_cluster) if args.csr else None,
                 "evals_per_step": evals / max(timed_iters, 1),
                 "data_passes_per_step": passes / max(timed_iters, 1),
