@@ -161,6 +161,9 @@ def main() -> int:
 
     comm = init_from_env()
     rank, world = comm.rank, comm.world_size
+    if args.gpus != world and rank == 0:
+        print(f"[bench] note: --gpus {args.gpus} but WORLD_SIZE={world}; "
+              "the record reports the ACTUAL world size", file=sys.stderr)
     if torch.cuda.is_available():
         device = torch.device("cuda", torch.cuda.current_device())
         dtype = {"bf16": torch.bfloat16, "f32": torch.float32,
