@@ -329,3 +329,20 @@ def test_gram_alternate_backtracking_fused_gpu():
     num = float(torch.norm(w_g - w_d))
     den = float(torch.norm(w_d)) + 1e-30
     assert num / den < 2e-2, (num, den)
+
+
+def test_gram_beta_ge_1_fused_gpu():
+    """beta >= 1 (no backtracking) through the FUSED gram trial on GPU:
+    exercises the short packed-read branch (lc_y + G row only, no x-loss)."""
+    shard, _ = generate_dense_problem(n=4096, d=16384, seed=91, device=DEV,
+                                      dtype=torch.bfloat16)
+    feats = shard.features.float()
+    Lsafe = float(0.25 * (feats * feats).sum() / shard.n) * 4.0
+    w0 = torch.zeros(16384, device=DEV, dtype=torch.float32)
+    args = (shard, LogisticGradient(), SimpleUpdater(), 1e-12, 10, 0.0, w0,
+            Lsafe, Lsafe, 1.5, 1.0, True)
+    w_d, h_d = run(*args, loss_history_mode="none")
+    w_g, h_g = run(*args, solver="gram", loss_history_mode="none")
+    assert len(h_d) == len(h_g)
+    for a, b in zip(h_d, h_g):
+        assert abs(a - b) < 5e-3 * max(1.0, abs(b)), (a, b)
