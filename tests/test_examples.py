@@ -38,3 +38,13 @@ def test_public_api_exports_resolve():
     assert m.__version__
     missing = [n for n in m.__all__ if not hasattr(m, n)]
     assert not missing, missing
+
+
+def test_iters_to_eps_script_runs():
+    """The iteration-advantage script (the metric's second half) stays
+    runnable; on CPU it uses its small fallback problem."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "benchmarks", "iters_to_eps.py")],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert out.returncode == 0, out.stderr[-2000:]
+    assert "AGD" in out.stdout and "L*" in out.stdout
