@@ -282,3 +282,36 @@ def test_backtrack_tol_config_is_honored():
                          loss_history_mode="backtrack"))
     opt.optimize(data, w0)
     assert g_cfg.loss_only <= 1
+
+
+def test_degenerate_inputs_are_safe():
+    """Degenerate shapes and data must terminate cleanly, not hang or NaN:
+    all-zero features (zero gradient -> exact-convergence break), a single
+    example, d=1, and constant labels."""
+    import math as _math
+
+    from sparkagd_amd.data import DenseShard
+
+    # all-zero features: gradient is identically zero
+    z = DenseShard(torch.zeros(50, 3, dtype=torch.float64), torch.ones(50))
+    w, h = run(z, LogisticGradient(), SimpleUpdater(), 1e-8, 20, 0.0,
+               torch.zeros(3, dtype=torch.float64), 1.0, _math.inf, 0.5,
+               0.9, True)
+    assert torch.all(w == 0) and len(h) >= 1
+    assert all(_math.isfinite(v) for v in h)
+
+    # one example, d=1
+    one = DenseShard(torch.tensor([[2.0]], dtype=torch.float64),
+                     torch.ones(1))
+    w, h = run(one, LogisticGradient(), SimpleUpdater(), 1e-10, 15, 0.0,
+               torch.zeros(1, dtype=torch.float64), 1.0, _math.inf, 0.5,
+               0.9, True)
+    assert _math.isfinite(float(w[0])) and h[-1] <= h[0]
+
+    # constant labels (all positive): perfectly separable in 1 direction
+    const = DenseShard(torch.randn(200, 4, dtype=torch.float64),
+                       torch.ones(200))
+    w, h = run(const, LogisticGradient(), SimpleUpdater(), 1e-10, 10, 0.0,
+               torch.zeros(4, dtype=torch.float64), 1.0, _math.inf, 0.5,
+               0.9, True)
+    assert all(_math.isfinite(v) for v in h) and h[-1] <= h[0]
